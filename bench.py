@@ -1,0 +1,145 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 ImageNet-shape data-parallel training.
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches this under torch.distributed.run (one rank per
+GPU over RCCL). Rank 0 prints ONE JSON line with the whole-job aggregate
+images/sec (BASELINE.json metric: "images/sec (whole node) ResNet-50
+ImageNet-shape at 1/2/4/8 MI355X").
+
+Timed region: K full training steps (forward, loss, backward with
+overlapped bucketed all-reduce, fused SGD update), bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX elapsed over ranks.
+Synthetic ImageNet-shape data (3x224x224), random-init weights, bf16
+compute with fp32 optimizer state.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=256,
+                   help="per-GPU batch size")
+    p.add_argument("--model", default="resnet50")
+    p.add_argument("--dtype", default="bfloat16",
+                   choices=["bfloat16", "float32"])
+    p.add_argument("--device", default="auto",
+                   help="auto|cuda|cpu (cpu only for plumbing tests)")
+    p.add_argument("--kernel-backend", default="auto",
+                   choices=["auto", "native", "torch"])
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from ddlbench_amd.config import BenchConfig
+    from ddlbench_amd.engine import compute_dtype, resolve_device
+    from ddlbench_amd.models import build_model
+    from ddlbench_amd.ops import functional as NF
+    from ddlbench_amd.ops.modules import set_default_backend
+    from ddlbench_amd.ops.sgd import FusedSGD
+    from ddlbench_amd.parallel import BucketedDataParallel, init_distributed
+
+    env = init_distributed()
+    world = env.world_size
+    cfg = BenchConfig(dataset="imagenet", arch=args.model, strategy="ddp",
+                      batch_size=args.batch, dtype=args.dtype,
+                      kernel_backend=args.kernel_backend, num_workers=0)
+    set_default_backend(cfg.kernel_backend)
+    device = resolve_device(cfg, env.local_rank)
+    dtype = compute_dtype(cfg)
+    if device.type == "cpu":
+        dtype = torch.float32  # plumbing mode only
+
+    torch.manual_seed(1234 + env.rank)
+    model = build_model(cfg.dataset, cfg.arch).to(device)
+    if dtype != torch.float32:
+        model = model.to(dtype)
+    dp = BucketedDataParallel(model)
+    opt = FusedSGD(model.parameters(), lr=0.1 * world, momentum=0.9,
+                   weight_decay=1e-4, backend=cfg.kernel_backend)
+
+    # synthetic device-resident batches, rotated so no step reuses the
+    # previous step's input (no caching of outputs; all compute runs)
+    n_pool = 4
+    pool = []
+    g = torch.Generator().manual_seed(7 + env.rank)
+    for _ in range(n_pool):
+        x = torch.randn(args.batch, 3, 224, 224, generator=g)
+        y = torch.randint(1000, (args.batch,), generator=g)
+        pool.append((x.to(device, dtype=dtype), y.to(device)))
+
+    def step(i):
+        x, y = pool[i % n_pool]
+        out = dp(x)
+        loss = NF.cross_entropy(out, y, backend=cfg.kernel_backend)
+        dp.zero_grad_buckets()
+        loss.backward()
+        dp.finalize_backward()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        loss = step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device.type == "cuda" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    if env.rank == 0:
+        total_images = args.steps * args.batch * world
+        value = total_images / elapsed
+        print(json.dumps({
+            "metric": "images/sec",
+            "value": value,
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": str(dtype).replace("torch.", ""),
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "input": "3x224x224",
+                "parallelism": f"dp{world}",
+            },
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

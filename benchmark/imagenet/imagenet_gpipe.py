@@ -1,0 +1,13 @@
+#!/usr/bin/env python3
+"""imagenet x gpipe benchmark entrypoint.
+
+MI355X-native counterpart of the reference's benchmark/imagenet/imagenet_gpipe.py
+(env contract: DATADIR/EPOCHS/BATCH_SIZE/LOGINTER/CORES_GPU[/MICROBATCHES];
+flags: -a/--arch, -s/--synthetic_data, --lr, --momentum). Strategy: gpipe.
+"""
+import os, sys
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", ".."))
+from ddlbench_amd.cli import main
+
+if __name__ == "__main__":
+    main("imagenet", "gpipe")

@@ -1,0 +1,38 @@
+"""Model zoo registry.
+
+One parametric implementation per family replaces the reference's three
+parallel per-dataset copies plus the per-dataset gpipemodels rebuilds
+(SURVEY.md §2.6): ``build_model(dataset, arch)`` resolves in_channels /
+num_classes / stem from the dataset, and every model exposes
+``to_sequential()`` for the pipeline engines.
+"""
+
+from __future__ import annotations
+
+import torch.nn as nn
+
+from ddlbench_amd.config import DATASET_SHAPES
+from ddlbench_amd.models.resnet import ResNet
+from ddlbench_amd.models.vgg import VGG
+from ddlbench_amd.models.mobilenetv2 import MobileNetV2
+
+RESNETS = ("resnet18", "resnet34", "resnet50", "resnet101", "resnet152")
+VGGS = ("vgg11", "vgg13", "vgg16", "vgg19")
+ARCHS = RESNETS + VGGS + ("mobilenetv2",)
+
+
+def build_model(dataset: str, arch: str) -> nn.Module:
+    c, h, w, ncls, _, _ = DATASET_SHAPES[dataset]
+    stem = "imagenet" if dataset in ("imagenet", "highres") else "small"
+    if arch in RESNETS:
+        return ResNet(arch, in_channels=c, num_classes=ncls, stem=stem)
+    if arch in VGGS:
+        return VGG(arch, in_channels=c, num_classes=ncls, stem=stem)
+    if arch == "mobilenetv2":
+        return MobileNetV2(in_channels=c, num_classes=ncls, stem=stem)
+    raise ValueError(f"unknown arch {arch!r}; choose from {ARCHS}")
+
+
+def build_sequential(dataset: str, arch: str) -> nn.Sequential:
+    """Sequential-flattened variant for pipeline partitioning."""
+    return build_model(dataset, arch).to_sequential()

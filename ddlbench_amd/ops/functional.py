@@ -1,0 +1,129 @@
+"""Autograd bridges for the native CDNA4 kernels, with plain-PyTorch
+reference implementations (the CPU path and the numerics-test oracle).
+
+Dispatch: ``ddlbench_amd.ops.use_native`` — native on HIP devices (hard
+error there if the extension is missing), torch composition on CPU."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from ddlbench_amd import ops as _ops
+
+_ACT = {"none": 0, "relu": 1, "relu6": 2}
+
+
+def _apply_act(v: torch.Tensor, act: str) -> torch.Tensor:
+    if act == "relu":
+        return F.relu(v, inplace=True)
+    if act == "relu6":
+        return F.relu6(v, inplace=True)
+    return v
+
+
+# ---------------------------------------------------------------- BN+act
+class _FusedBNAct(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, res, running_mean, running_var,
+                training, momentum, eps, act_code):
+        ext = _ops.require_extension()
+        x = x.contiguous()
+        res = res.contiguous() if res is not None else None
+        y, mean, invstd = ext.bn_act_fwd(x, res, gamma, beta, running_mean,
+                                         running_var, training, momentum,
+                                         eps, act_code)
+        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        ctx.training = training
+        ctx.act_code = act_code
+        ctx.has_res = res is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _ops.require_extension()
+        x, y, mean, invstd, gamma = ctx.saved_tensors
+        out = ext.bn_act_bwd(dy.contiguous(), y, x, mean, invstd, gamma,
+                             ctx.act_code, ctx.training, ctx.has_res)
+        dx, dgamma, dbeta = out[0], out[1], out[2]
+        dres = out[3] if ctx.has_res else None
+        return (dx, dgamma, dbeta, dres, None, None, None, None, None, None)
+
+
+def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+           running_mean: Optional[torch.Tensor],
+           running_var: Optional[torch.Tensor], training: bool,
+           momentum: float = 0.1, eps: float = 1e-5, act: str = "relu",
+           res: Optional[torch.Tensor] = None,
+           backend: str = "auto") -> torch.Tensor:
+    """BatchNorm2d + activation (+ residual add), fused on GPU."""
+    if _ops.use_native(x, backend):
+        return _FusedBNAct.apply(x, gamma.float(), beta.float(), res,
+                                 running_mean, running_var, training,
+                                 momentum, eps, _ACT[act])
+    y = F.batch_norm(x, running_mean, running_var, gamma, beta, training,
+                     momentum, eps)
+    if res is not None:
+        y = y + res
+    return _apply_act(y, act)
+
+
+# ----------------------------------------------------------- cross entropy
+class _FusedCrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        ext = _ops.require_extension()
+        logits = logits.contiguous()
+        loss_sum, lse = ext.ce_fwd(logits, target)
+        ctx.save_for_backward(logits, target, lse)
+        return (loss_sum / logits.size(0)).squeeze(0)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        ext = _ops.require_extension()
+        logits, target, lse = ctx.saved_tensors
+        gscale = (grad_out.detach().float() / logits.size(0)).reshape(1)
+        dx = ext.ce_bwd(logits, target, lse, gscale.contiguous())
+        return dx, None
+
+
+def cross_entropy(logits: torch.Tensor, target: torch.Tensor,
+                  backend: str = "auto") -> torch.Tensor:
+    if _ops.use_native(logits, backend):
+        return _FusedCrossEntropy.apply(logits, target)
+    return F.cross_entropy(logits, target)
+
+
+# ------------------------------------------------------- depthwise conv3x3
+class _FusedDWConv3x3(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        ext = _ops.require_extension()
+        x = x.contiguous()
+        w32 = weight.detach().float().contiguous()
+        y = ext.dw3x3_fwd(x, w32, stride)
+        ctx.save_for_backward(x, w32)
+        ctx.stride = stride
+        ctx.wdtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _ops.require_extension()
+        x, w32 = ctx.saved_tensors
+        need_dx, need_dw = ctx.needs_input_grad[0], ctx.needs_input_grad[1]
+        dx, dw = ext.dw3x3_bwd(x, w32, dy.contiguous(), ctx.stride,
+                               need_dx, need_dw)
+        if need_dw:
+            dw = dw.to(ctx.wdtype)
+        return (dx if need_dx else None), (dw if need_dw else None), None
+
+
+def depthwise_conv3x3(x: torch.Tensor, weight: torch.Tensor,
+                      stride: int = 1, backend: str = "auto") -> torch.Tensor:
+    """3x3 depthwise conv, pad 1, groups == channels."""
+    if _ops.use_native(x, backend):
+        return _FusedDWConv3x3.apply(x, weight, stride)
+    return F.conv2d(x, weight, None, stride, 1, 1, groups=x.size(1))

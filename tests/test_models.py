@@ -1,0 +1,62 @@
+"""Model zoo: shapes, backward, sequential equivalence."""
+
+import pytest
+import torch
+
+from ddlbench_amd.config import DATASET_SHAPES
+from ddlbench_amd.models import ARCHS, build_model, build_sequential
+
+CASES = [
+    ("mnist", "resnet18"), ("mnist", "vgg11"), ("mnist", "mobilenetv2"),
+    ("cifar10", "resnet50"), ("cifar10", "vgg16"),
+    ("cifar10", "mobilenetv2"), ("imagenet", "resnet50"),
+    ("imagenet", "vgg16"), ("imagenet", "mobilenetv2"),
+]
+
+
+@pytest.mark.parametrize("dataset,arch", CASES)
+def test_forward_backward_shapes(dataset, arch):
+    torch.manual_seed(0)
+    c, h, w, ncls, _, _ = DATASET_SHAPES[dataset]
+    m = build_model(dataset, arch)
+    x = torch.randn(2, c, h, w)
+    y = m(x)
+    assert y.shape == (2, ncls)
+    y.sum().backward()
+    grads = [p.grad for p in m.parameters() if p.requires_grad]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+@pytest.mark.parametrize("dataset,arch",
+                         [("mnist", "resnet18"), ("cifar10", "mobilenetv2"),
+                          ("cifar10", "vgg11")])
+def test_sequential_matches_model(dataset, arch):
+    """to_sequential() must compute the same function."""
+    torch.manual_seed(0)
+    c, h, w, ncls, _, _ = DATASET_SHAPES[dataset]
+    m = build_model(dataset, arch).eval()
+    seq = m.to_sequential().eval()
+    x = torch.randn(2, c, h, w)
+    with torch.no_grad():
+        torch.testing.assert_close(m(x), seq(x), rtol=1e-5, atol=1e-5)
+
+
+def test_all_archs_instantiate():
+    for arch in ARCHS:
+        m = build_model("cifar10", arch)
+        assert sum(p.numel() for p in m.parameters()) > 0
+
+
+def test_resnet50_param_count_matches_torchvision():
+    """25.557M params — parity check vs the reference's torchvision
+    resnet50 (imagenet_pytorch.py:19-30)."""
+    m = build_model("imagenet", "resnet50")
+    assert sum(p.numel() for p in m.parameters()) == 25_557_032
+
+
+def test_highres_shape_forward():
+    m = build_model("highres", "resnet18").eval()
+    with torch.no_grad():
+        y = m(torch.randn(1, 3, 512, 512))
+    assert y.shape == (1, 1000)
