@@ -42,6 +42,11 @@ def parse_args():
                    help="auto|cuda|cpu (cpu only for plumbing tests)")
     p.add_argument("--kernel-backend", default="auto",
                    choices=["auto", "native", "torch"])
+    p.add_argument("--memory-format", default="channels_last",
+                   choices=["channels_last", "contiguous"],
+                   help="channels_last (NHWC) keeps MIOpen on its native "
+                        "xdlops conv solvers and our BN kernels on the "
+                        "coalesced NHWC path")
     return p.parse_args()
 
 
@@ -66,10 +71,15 @@ def main():
     if device.type == "cpu":
         dtype = torch.float32  # plumbing mode only
 
+    torch.backends.cudnn.benchmark = True  # MIOpen find for best solvers
+    channels_last = (args.memory_format == "channels_last"
+                     and device.type == "cuda")
     torch.manual_seed(1234 + env.rank)
     model = build_model(cfg.dataset, cfg.arch).to(device)
     if dtype != torch.float32:
         model = model.to(dtype)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     dp = BucketedDataParallel(model)
     opt = FusedSGD(model.parameters(), lr=0.1 * world, momentum=0.9,
                    weight_decay=1e-4, backend=cfg.kernel_backend)
@@ -82,7 +92,10 @@ def main():
     for _ in range(n_pool):
         x = torch.randn(args.batch, 3, 224, 224, generator=g)
         y = torch.randint(1000, (args.batch,), generator=g)
-        pool.append((x.to(device, dtype=dtype), y.to(device)))
+        x = x.to(device, dtype=dtype)
+        if channels_last:
+            x = x.contiguous(memory_format=torch.channels_last)
+        pool.append((x, y.to(device)))
 
     def step(i):
         x, y = pool[i % n_pool]

@@ -16,25 +16,25 @@ void launch_fused_sgd(uintptr_t*, uintptr_t*, uintptr_t*, const int64_t*,
                       int, int64_t, float, float, float, int, hipStream_t);
 
 template <typename T>
-void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t,
+void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int,
                      hipStream_t);
 void launch_bn_finalize(const double*, float*, float*, float*, float*,
                         int64_t, double, float, float, hipStream_t);
 template <typename T>
 void launch_bn_apply(const T*, const T*, T*, const float*, const float*,
                      const float*, const float*, int64_t, int64_t, int64_t,
-                     int, hipStream_t);
+                     int, int, hipStream_t);
 template <typename T>
 void launch_bn_bwd_reduce(const T*, const T*, const T*, const float*,
                           const float*, double*, int64_t, int64_t, int64_t,
-                          int, hipStream_t);
+                          int, int, hipStream_t);
 void launch_bn_bwd_finalize(const double*, const float*, const float*,
                             float*, float*, float*, int64_t, double, int,
                             hipStream_t);
 template <typename T>
 void launch_bn_bwd_dx(const T*, const T*, const T*, const float*,
                       const float*, const float*, T*, T*, int64_t, int64_t,
-                      int64_t, int, hipStream_t);
+                      int64_t, int, int, hipStream_t);
 
 template <typename T>
 void launch_ce_fwd(const T*, const int64_t*, float*, float*, int64_t,
@@ -103,9 +103,11 @@ std::vector<torch::Tensor> bn_act_fwd(
     torch::Tensor x, c10::optional<torch::Tensor> res, torch::Tensor gamma,
     torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
     c10::optional<torch::Tensor> running_var, bool training, double momentum,
-    double eps, int64_t act) {
-  check_gpu_contig(x, "x");
-  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+    double eps, int64_t act, bool nhwc) {
+  TORCH_CHECK(x.is_cuda(), "x must be on the HIP device");
+  TORCH_CHECK(x.dim() == 4, "x must be 4-D");
+  TORCH_CHECK(nhwc ? x.is_contiguous(at::MemoryFormat::ChannelsLast)
+                   : x.is_contiguous(), "x layout mismatch");
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto s = cur_stream();
   auto fopt = x.options().dtype(at::kFloat);
@@ -115,9 +117,10 @@ std::vector<torch::Tensor> bn_act_fwd(
     torch::Tensor sums = torch::zeros({2, C}, x.options().dtype(at::kDouble));
     if (is_bf16(x))
       launch_bn_stats<__hip_bfloat16>(dptr<__hip_bfloat16>(x),
-                                      dptr<double>(sums), N, C, HW, s);
+                                      dptr<double>(sums), N, C, HW, nhwc, s);
     else
-      launch_bn_stats<float>(dptr<float>(x), dptr<double>(sums), N, C, HW, s);
+      launch_bn_stats<float>(dptr<float>(x), dptr<double>(sums), N, C, HW,
+                             nhwc, s);
     launch_bn_finalize(
         dptr<double>(sums), dptr<float>(mean), dptr<float>(invstd),
         running_mean ? dptr<float>(*running_mean) : nullptr,
@@ -136,13 +139,13 @@ std::vector<torch::Tensor> bn_act_fwd(
         dptr<__hip_bfloat16>(x),
         res ? dptr<__hip_bfloat16>(*res) : nullptr, dptr<__hip_bfloat16>(y),
         dptr<float>(mean), dptr<float>(invstd), dptr<float>(gamma),
-        dptr<float>(beta), C, HW, total, (int)act, s);
+        dptr<float>(beta), C, HW, total, (int)act, nhwc, s);
   else
     launch_bn_apply<float>(dptr<float>(x),
                            res ? dptr<float>(*res) : nullptr, dptr<float>(y),
                            dptr<float>(mean), dptr<float>(invstd),
                            dptr<float>(gamma), dptr<float>(beta), C, HW,
-                           total, (int)act, s);
+                           total, (int)act, nhwc, s);
   return {y, mean, invstd};
 }
 
@@ -150,8 +153,9 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
                                       torch::Tensor x, torch::Tensor mean,
                                       torch::Tensor invstd,
                                       torch::Tensor gamma, int64_t act,
-                                      bool training, bool need_dres) {
-  check_gpu_contig(dy, "dy");
+                                      bool training, bool need_dres,
+                                      bool nhwc) {
+  TORCH_CHECK(dy.is_cuda(), "dy must be on the HIP device");
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto s = cur_stream();
   auto fopt = x.options().dtype(at::kFloat);
@@ -167,7 +171,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
     launch_bn_bwd_reduce<__hip_bfloat16>(
         dptr<__hip_bfloat16>(dy), dptr<__hip_bfloat16>(y),
         dptr<__hip_bfloat16>(x), dptr<float>(mean), dptr<float>(invstd),
-        dptr<double>(sums), N, C, HW, (int)act, s);
+        dptr<double>(sums), N, C, HW, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
                            dptr<float>(dbeta), dptr<float>(k), C,
@@ -177,12 +181,12 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
         dptr<__hip_bfloat16>(x), dptr<float>(mean), dptr<float>(invstd),
         dptr<float>(k), dptr<__hip_bfloat16>(dx),
         need_dres ? dptr<__hip_bfloat16>(dres) : nullptr, C, HW, total,
-        (int)act, s);
+        (int)act, nhwc, s);
   } else {
     launch_bn_bwd_reduce<float>(dptr<float>(dy), dptr<float>(y),
                                 dptr<float>(x), dptr<float>(mean),
                                 dptr<float>(invstd), dptr<double>(sums), N,
-                                C, HW, (int)act, s);
+                                C, HW, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
                            dptr<float>(dbeta), dptr<float>(k), C,
@@ -191,7 +195,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
                             dptr<float>(mean), dptr<float>(invstd),
                             dptr<float>(k), dptr<float>(dx),
                             need_dres ? dptr<float>(dres) : nullptr, C, HW,
-                            total, (int)act, s);
+                            total, (int)act, nhwc, s);
   }
   if (need_dres) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};

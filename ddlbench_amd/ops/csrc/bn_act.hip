@@ -63,6 +63,28 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, double* __restrict__ su
   }
 }
 
+// ---- stats, NHWC (channels-last): thread-per-channel, coalesced -------
+// rows = N*H*W; element (row, c) at row*C + c. Consecutive lanes read
+// consecutive channels -> every row access is one coalesced segment.
+template <typename T>
+__global__ void bn_stats_nhwc_kernel(const T* __restrict__ x,
+                                     double* __restrict__ sums,
+                                     int64_t rows, int64_t C) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  double s = 0.0, ss = 0.0;
+  for (int64_t r = begin; r < end; ++r) {
+    const float v = to_f32(x[r * C + c]);
+    s += v;
+    ss += fma((double)v, (double)v, 0.0);
+  }
+  atomicAdd(&sums[c], s);
+  atomicAdd(&sums[C + c], ss);
+}
+
 // ---- finalize: mean/invstd + running-stat update ----------------------
 __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ mean,
@@ -94,11 +116,11 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                int64_t C, int64_t HW, int64_t total) {
+                                int64_t C, int64_t cdiv, int64_t total) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    const int64_t c = (i / HW) % C;
+    const int64_t c = (i / cdiv) % C;
     float v = (to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (ADD) v += to_f32(res[i]);
     y[i] = from_f32<T>(act_fwd<ACT>(v));
@@ -140,6 +162,33 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
   }
 }
 
+// ---- backward reduce, NHWC: thread-per-channel, coalesced -------------
+template <typename T, int ACT>
+__global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
+                                          const T* __restrict__ y,
+                                          const T* __restrict__ x,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          double* __restrict__ sums,
+                                          int64_t rows, int64_t C) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mu = mean[c], is = invstd[c];
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  double sdy = 0.0, sdyx = 0.0;
+  for (int64_t r = begin; r < end; ++r) {
+    const int64_t idx = r * C + c;
+    const float g = to_f32(dy[idx]) * act_mask<ACT>(to_f32(y[idx]));
+    const float xhat = (to_f32(x[idx]) - mu) * is;
+    sdy += g;
+    sdyx += fma((double)g, (double)xhat, 0.0);
+  }
+  atomicAdd(&sums[c], sdy);
+  atomicAdd(&sums[C + c], sdyx);
+}
+
 __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ invstd,
@@ -167,11 +216,11 @@ __global__ void bn_bwd_dx_kernel(const T* __restrict__ dy,
                                  const float* __restrict__ invstd,
                                  const float* __restrict__ k,
                                  T* __restrict__ dx, T* __restrict__ dres,
-                                 int64_t C, int64_t HW, int64_t total) {
+                                 int64_t C, int64_t cdiv, int64_t total) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    const int64_t c = (i / HW) % C;
+    const int64_t c = (i / cdiv) % C;
     const float g = to_f32(dy[i]) * act_mask<ACT>(to_f32(y[i]));
     const float xhat = (to_f32(x[i]) - mean[c]) * invstd[c];
     dx[i] = from_f32<T>(k[c] * (g - k[C + c] - xhat * k[2 * C + c]));
@@ -187,13 +236,20 @@ static inline int elementwise_grid(int64_t total, int block) {
 
 template <typename T>
 void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
-                     int64_t HW, hipStream_t stream) {
+                     int64_t HW, int nhwc, hipStream_t stream) {
   const int block = 256;
-  int64_t S = i64min((N * HW + block - 1) / block,
-                           i64max(2048 / C, 1));
-  S = i64max(S, 1);
-  hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(C, S), dim3(block), 0,
-                     stream, x, sums, N, C, HW);
+  if (nhwc) {
+    const int64_t rows = N * HW;
+    const int64_t cblocks = (C + block - 1) / block;
+    int64_t S = i64min(i64max(rows / 1024, 1), i64max(2048 / cblocks, 1));
+    hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
+                       dim3(block), 0, stream, x, sums, rows, C);
+  } else {
+    int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
+    S = i64max(S, 1);
+    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(C, S), dim3(block), 0,
+                       stream, x, sums, N, C, HW);
+  }
   HIP_CHECK_LAST();
 }
 
@@ -211,13 +267,14 @@ template <typename T>
 void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
                      const float* invstd, const float* gamma,
                      const float* beta, int64_t C, int64_t HW, int64_t total,
-                     int act, hipStream_t stream) {
+                     int act, int nhwc, hipStream_t stream) {
   const int block = 256;
   const int grid = elementwise_grid(total, block);
+  const int64_t cdiv = nhwc ? 1 : HW;
 #define CASE(ACT, ADD)                                                       \
   hipLaunchKernelGGL((bn_apply_kernel<T, ACT, ADD>), dim3(grid), dim3(block), \
-                     0, stream, x, res, y, mean, invstd, gamma, beta, C, HW, \
-                     total)
+                     0, stream, x, res, y, mean, invstd, gamma, beta, C,    \
+                     cdiv, total)
   const bool add = res != nullptr;
   if (act == 0 && !add) CASE(0, false);
   else if (act == 0 && add) CASE(0, true);
@@ -233,19 +290,32 @@ template <typename T>
 void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
                           const float* mean, const float* invstd,
                           double* sums, int64_t N, int64_t C, int64_t HW,
-                          int act, hipStream_t stream) {
+                          int act, int nhwc, hipStream_t stream) {
   const int block = 256;
-  int64_t S = i64min((N * HW + block - 1) / block,
-                           i64max(2048 / C, 1));
-  S = i64max(S, 1);
-#define CASE(ACT)                                                          \
-  hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, ACT>), dim3(C, S),           \
-                     dim3(block), 0, stream, dy, y, x, mean, invstd, sums, \
-                     N, C, HW)
-  if (act == 0) CASE(0);
-  else if (act == 1) CASE(1);
-  else CASE(2);
+  if (nhwc) {
+    const int64_t rows = N * HW;
+    const int64_t cblocks = (C + block - 1) / block;
+    int64_t S = i64min(i64max(rows / 1024, 1), i64max(2048 / cblocks, 1));
+#define CASE(ACT)                                                           \
+    hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT>),                 \
+                       dim3(cblocks, S), dim3(block), 0, stream, dy, y, x,  \
+                       mean, invstd, sums, rows, C)
+    if (act == 0) CASE(0);
+    else if (act == 1) CASE(1);
+    else CASE(2);
 #undef CASE
+  } else {
+    int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
+    S = i64max(S, 1);
+#define CASE(ACT)                                                          \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, ACT>), dim3(C, S),         \
+                       dim3(block), 0, stream, dy, y, x, mean, invstd,     \
+                       sums, N, C, HW)
+    if (act == 0) CASE(0);
+    else if (act == 1) CASE(1);
+    else CASE(2);
+#undef CASE
+  }
   HIP_CHECK_LAST();
 }
 
@@ -264,13 +334,14 @@ template <typename T>
 void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
                       const float* invstd, const float* k, T* dx, T* dres,
                       int64_t C, int64_t HW, int64_t total, int act,
-                      hipStream_t stream) {
+                      int nhwc, hipStream_t stream) {
   const int block = 256;
   const int grid = elementwise_grid(total, block);
+  const int64_t cdiv = nhwc ? 1 : HW;
 #define CASE(ACT, ADD)                                                    \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<T, ACT, ADD>), dim3(grid),         \
                      dim3(block), 0, stream, dy, y, x, mean, invstd, k,   \
-                     dx, dres, C, HW, total)
+                     dx, dres, C, cdiv, total)
   const bool add = dres != nullptr;
   if (act == 0 && !add) CASE(0, false);
   else if (act == 0 && add) CASE(0, true);
@@ -284,19 +355,19 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
 
 #define INSTANTIATE(T)                                                        \
   template void launch_bn_stats<T>(const T*, double*, int64_t, int64_t,       \
-                                   int64_t, hipStream_t);                     \
+                                   int64_t, int, hipStream_t);                \
   template void launch_bn_apply<T>(const T*, const T*, T*, const float*,      \
                                    const float*, const float*, const float*,  \
-                                   int64_t, int64_t, int64_t, int,            \
+                                   int64_t, int64_t, int64_t, int, int,       \
                                    hipStream_t);                              \
   template void launch_bn_bwd_reduce<T>(const T*, const T*, const T*,         \
                                         const float*, const float*, double*,  \
-                                        int64_t, int64_t, int64_t, int,       \
+                                        int64_t, int64_t, int64_t, int, int,  \
                                         hipStream_t);                         \
   template void launch_bn_bwd_dx<T>(const T*, const T*, const T*,             \
                                     const float*, const float*, const float*, \
                                     T*, T*, int64_t, int64_t, int64_t, int,   \
-                                    hipStream_t);
+                                    int, hipStream_t);
 
 INSTANTIATE(float)
 INSTANTIATE(__hip_bfloat16)

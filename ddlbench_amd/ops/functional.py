@@ -16,6 +16,18 @@ from ddlbench_amd import ops as _ops
 _ACT = {"none": 0, "relu": 1, "relu6": 2}
 
 
+def _is_nhwc(t: torch.Tensor) -> bool:
+    return (t.dim() == 4
+            and t.is_contiguous(memory_format=torch.channels_last)
+            and not t.is_contiguous())
+
+
+def _layout_contiguous(t: torch.Tensor, nhwc: bool) -> torch.Tensor:
+    if nhwc:
+        return t.contiguous(memory_format=torch.channels_last)
+    return t.contiguous()
+
+
 def _apply_act(v: torch.Tensor, act: str) -> torch.Tensor:
     if act == "relu":
         return F.relu(v, inplace=True)
@@ -30,23 +42,26 @@ class _FusedBNAct(torch.autograd.Function):
     def forward(ctx, x, gamma, beta, res, running_mean, running_var,
                 training, momentum, eps, act_code):
         ext = _ops.require_extension()
-        x = x.contiguous()
-        res = res.contiguous() if res is not None else None
+        nhwc = _is_nhwc(x)
+        x = _layout_contiguous(x, nhwc)
+        res = _layout_contiguous(res, nhwc) if res is not None else None
         y, mean, invstd = ext.bn_act_fwd(x, res, gamma, beta, running_mean,
                                          running_var, training, momentum,
-                                         eps, act_code)
+                                         eps, act_code, nhwc)
         ctx.save_for_backward(x, y, mean, invstd, gamma)
         ctx.training = training
         ctx.act_code = act_code
         ctx.has_res = res is not None
+        ctx.nhwc = nhwc
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = _ops.require_extension()
         x, y, mean, invstd, gamma = ctx.saved_tensors
-        out = ext.bn_act_bwd(dy.contiguous(), y, x, mean, invstd, gamma,
-                             ctx.act_code, ctx.training, ctx.has_res)
+        out = ext.bn_act_bwd(_layout_contiguous(dy, ctx.nhwc), y, x, mean,
+                             invstd, gamma, ctx.act_code, ctx.training,
+                             ctx.has_res, ctx.nhwc)
         dx, dgamma, dbeta = out[0], out[1], out[2]
         dres = out[3] if ctx.has_res else None
         return (dx, dgamma, dbeta, dres, None, None, None, None, None, None)

@@ -1,0 +1,213 @@
+"""Layer-graph IR shared by the profiler and the partitioner.
+
+Our rebuild of the reference's graph library
+(/root/reference/pipedream-fork/graph/graph.py: Node with fwd/bwd
+compute_time, activation_size, parameter_size, stage_id; topological sort,
+predecessors/successors, antichain DAG, partition_graph, text
+serialization — SURVEY.md §2.7). The algorithms are re-implemented from
+the problem statement, not translated; the text format is ours (one
+node/edge per line, round-trippable).
+"""
+
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Set, Tuple
+
+
+@dataclass
+class Node:
+    node_id: int
+    desc: str = ""
+    fwd_time: float = 0.0          # seconds
+    bwd_time: float = 0.0          # seconds
+    activation_size: float = 0.0   # bytes (output)
+    parameter_size: float = 0.0    # bytes
+    stage_id: int = -1
+
+    @property
+    def compute_time(self) -> float:
+        return self.fwd_time + self.bwd_time
+
+
+class Graph:
+    def __init__(self) -> None:
+        self.nodes: Dict[int, Node] = {}
+        self.edges: Dict[int, List[int]] = {}      # src -> [dst]
+        self.in_edges: Dict[int, List[int]] = {}   # dst -> [src]
+
+    # ---- construction --------------------------------------------------
+    def add_node(self, node: Node) -> Node:
+        self.nodes[node.node_id] = node
+        self.edges.setdefault(node.node_id, [])
+        self.in_edges.setdefault(node.node_id, [])
+        return node
+
+    def add_edge(self, src: int, dst: int) -> None:
+        if dst not in self.edges[src]:
+            self.edges[src].append(dst)
+            self.in_edges[dst].append(src)
+
+    @classmethod
+    def chain(cls, nodes: List[Node]) -> "Graph":
+        g = cls()
+        for n in nodes:
+            g.add_node(n)
+        for a, b in zip(nodes, nodes[1:]):
+            g.add_edge(a.node_id, b.node_id)
+        return g
+
+    # ---- queries -------------------------------------------------------
+    def sources(self) -> List[Node]:
+        return [self.nodes[i] for i in self.nodes if not self.in_edges[i]]
+
+    def sinks(self) -> List[Node]:
+        return [self.nodes[i] for i in self.nodes if not self.edges[i]]
+
+    def topological_sort(self) -> List[Node]:
+        indeg = {i: len(self.in_edges[i]) for i in self.nodes}
+        ready = sorted(i for i, d in indeg.items() if d == 0)
+        order = []
+        while ready:
+            i = ready.pop(0)
+            order.append(self.nodes[i])
+            for j in self.edges[i]:
+                indeg[j] -= 1
+                if indeg[j] == 0:
+                    ready.append(j)
+            ready.sort()
+        if len(order) != len(self.nodes):
+            raise ValueError("graph has a cycle")
+        return order
+
+    def predecessors(self, node_id: int) -> Set[int]:
+        """All transitive predecessors (memoization-free DFS)."""
+        seen: Set[int] = set()
+        stack = list(self.in_edges[node_id])
+        while stack:
+            i = stack.pop()
+            if i not in seen:
+                seen.add(i)
+                stack.extend(self.in_edges[i])
+        return seen
+
+    def successors(self, node_id: int) -> Set[int]:
+        seen: Set[int] = set()
+        stack = list(self.edges[node_id])
+        while stack:
+            i = stack.pop()
+            if i not in seen:
+                seen.add(i)
+                stack.extend(self.edges[i])
+        return seen
+
+    def is_chain(self) -> bool:
+        return all(len(v) <= 1 for v in self.edges.values()) and \
+            all(len(v) <= 1 for v in self.in_edges.values())
+
+    # ---- antichains (partition frontiers for non-chain DAGs) -----------
+    def antichain_dag(self) -> Tuple[List[frozenset], Dict[frozenset,
+                                                           List[frozenset]]]:
+        """Enumerate the cut frontiers of the DAG.
+
+        An antichain here is a minimal set of nodes whose removal (with
+        all their predecessors) splits the graph — the candidate pipeline
+        split points (the reference enumerates the same states,
+        graph.py:420-449). Returns (states in topo order of discovery,
+        adjacency). Exponential in width; our model graphs are chains or
+        near-chains so this stays tiny."""
+        start = frozenset(n.node_id for n in self.sources())
+        states: List[frozenset] = [start]
+        adj: Dict[frozenset, List[frozenset]] = {start: []}
+        work = [start]
+        while work:
+            ac = work.pop(0)
+            # set of nodes "covered" = antichain + its predecessors
+            covered: Set[int] = set(ac)
+            for i in ac:
+                covered |= self.predecessors(i)
+            for i in ac:
+                for j in self.edges[i]:
+                    # next antichain: advance over node j if all its
+                    # parents are covered
+                    if not all(p in covered or p == i
+                               for p in self.in_edges[j]):
+                        continue
+                    nxt = (set(ac) - {i}) | {j}
+                    # drop members that are now predecessors of others
+                    nxt_f = frozenset(
+                        m for m in nxt
+                        if not any(m in self.predecessors(o)
+                                   for o in nxt if o != m))
+                    if nxt_f not in adj:
+                        adj[nxt_f] = []
+                        states.append(nxt_f)
+                        work.append(nxt_f)
+                    if nxt_f not in adj[ac]:
+                        adj[ac].append(nxt_f)
+        return states, adj
+
+    # ---- partitioning ---------------------------------------------------
+    def partition_by_stage(self) -> Dict[int, "Graph"]:
+        """Split into per-stage subgraphs by node.stage_id
+        (reference partition_graph, graph.py:117-137)."""
+        out: Dict[int, Graph] = {}
+        for n in self.nodes.values():
+            g = out.setdefault(n.stage_id, Graph())
+            g.add_node(n)
+        for src, dsts in self.edges.items():
+            for dst in dsts:
+                s = self.nodes[src].stage_id
+                if s == self.nodes[dst].stage_id:
+                    out[s].add_edge(src, dst)
+        return out
+
+    def stage_boundaries(self) -> List[Tuple[int, int]]:
+        """(src, dst) edges that cross stages, in topo order."""
+        res = []
+        for n in self.topological_sort():
+            for dst in self.edges[n.node_id]:
+                if self.nodes[dst].stage_id != n.stage_id:
+                    res.append((n.node_id, dst))
+        return res
+
+    # ---- serialization (our own line format) ----------------------------
+    def dumps(self) -> str:
+        lines = []
+        for n in self.topological_sort():
+            lines.append("node " + json.dumps({
+                "id": n.node_id, "desc": n.desc, "fwd_time": n.fwd_time,
+                "bwd_time": n.bwd_time, "activation_size": n.activation_size,
+                "parameter_size": n.parameter_size, "stage_id": n.stage_id}))
+        for src in sorted(self.edges):
+            for dst in self.edges[src]:
+                lines.append(f"edge {src} {dst}")
+        return "\n".join(lines) + "\n"
+
+    @classmethod
+    def loads(cls, text: str) -> "Graph":
+        g = cls()
+        edges = []
+        for line in text.splitlines():
+            line = line.strip()
+            if line.startswith("node "):
+                d = json.loads(line[5:])
+                g.add_node(Node(d["id"], d["desc"], d["fwd_time"],
+                                d["bwd_time"], d["activation_size"],
+                                d["parameter_size"], d["stage_id"]))
+            elif line.startswith("edge "):
+                _, s, t = line.split()
+                edges.append((int(s), int(t)))
+        for s, t in edges:
+            g.add_edge(s, t)
+        return g
+
+    def save(self, path: str) -> None:
+        with open(path, "w") as f:
+            f.write(self.dumps())
+
+    @classmethod
+    def load(cls, path: str) -> "Graph":
+        with open(path) as f:
+            return cls.loads(f.read())
