@@ -127,25 +127,23 @@ class Graph:
             covered: Set[int] = set(ac)
             for i in ac:
                 covered |= self.predecessors(i)
-            for i in ac:
-                for j in self.edges[i]:
-                    # next antichain: advance over node j if all its
-                    # parents are covered
-                    if not all(p in covered or p == i
-                               for p in self.in_edges[j]):
-                        continue
-                    nxt = (set(ac) - {i}) | {j}
-                    # drop members that are now predecessors of others
-                    nxt_f = frozenset(
-                        m for m in nxt
-                        if not any(m in self.predecessors(o)
-                                   for o in nxt if o != m))
-                    if nxt_f not in adj:
-                        adj[nxt_f] = []
-                        states.append(nxt_f)
-                        work.append(nxt_f)
-                    if nxt_f not in adj[ac]:
-                        adj[ac].append(nxt_f)
+            # candidates: any uncovered node whose every parent is covered
+            candidates = {j for j in self.nodes
+                          if j not in covered and self.in_edges[j]
+                          and all(p in covered for p in self.in_edges[j])}
+            for j in candidates:
+                nxt = set(ac) | {j}
+                # keep only maximal members (drop predecessors of others)
+                nxt_f = frozenset(
+                    m for m in nxt
+                    if not any(m in self.predecessors(o)
+                               for o in nxt if o != m))
+                if nxt_f not in adj:
+                    adj[nxt_f] = []
+                    states.append(nxt_f)
+                    work.append(nxt_f)
+                if nxt_f not in adj[ac]:
+                    adj[ac].append(nxt_f)
         return states, adj
 
     # ---- partitioning ---------------------------------------------------

@@ -1,0 +1,250 @@
+"""PipeDream-style training driver: profile -> partition -> 1F1B runtime.
+
+One process per GPU (torchrun). Rank 0 profiles the layer chain and runs
+the partitioner; the plan broadcasts to all ranks (replacing the
+reference's offline profiler -> optimizer -> codegen -> bash-parsed
+stage map flow, SURVEY.md §3.3 — stages are sliced from the live module
+list at runtime; the conf.json contract is still written to disk).
+
+Epoch flow mirrors main_with_runtime.py:406-500: warmup forwards, 1F1B
+steady state with per-minibatch optimizer steps, backward drain,
+forward-only validation with ack clocking."""
+
+from __future__ import annotations
+
+import json
+import math
+import os
+import time
+from typing import List
+
+import torch
+import torch.distributed as dist
+
+from ddlbench_amd.config import BenchConfig
+from ddlbench_amd.data import SyntheticImageDataset
+from ddlbench_amd.engine import compute_dtype, resolve_device
+from ddlbench_amd.models import build_sequential
+from ddlbench_amd.ops import functional as NF
+from ddlbench_amd.ops.modules import set_default_backend
+from ddlbench_amd.ops.sgd import FusedSGD
+from ddlbench_amd.parallel import BucketedDataParallel, init_distributed
+from ddlbench_amd.parallel.dist_utils import distributed_env
+from ddlbench_amd.parallel.pipeline.comm import (PipelineTransport,
+                                                 dry_run_shapes)
+from ddlbench_amd.parallel.pipeline.graph import Graph
+from ddlbench_amd.parallel.pipeline.partition import (PartitionResult,
+                                                      partition_chain)
+from ddlbench_amd.parallel.pipeline.profiler import profile_sequential
+from ddlbench_amd.parallel.pipeline.runtime import StagePlan, StageRuntime
+from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
+from ddlbench_amd.utils import AverageMeter, BenchLogger, accuracy
+
+
+def _make_plan(cfg: BenchConfig, seq, device, world: int,
+               straight: bool = False):
+    """Rank 0 profiles + partitions; everyone gets the same plan."""
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    payload = [None]
+    if rank == 0:
+        from ddlbench_amd.data import synthetic_batch
+        sample, _ = synthetic_batch(cfg, batch_size=min(cfg.batch_size, 8),
+                                    device=device,
+                                    dtype=compute_dtype(cfg))
+        graph = profile_sequential(seq, sample, device=device,
+                                   iters=3, warmup=1)
+        result = partition_chain(graph, world, straight=straight)
+        os.makedirs(f"profiles/{cfg.arch}", exist_ok=True)
+        graph.save(f"profiles/{cfg.arch}/graph.txt")
+        result.save(f"profiles/{cfg.arch}/conf.json")
+        payload = [{
+            "module_to_stage_map": result.module_to_stage_map,
+            "replicas": [s.replicas for s in result.stages],
+        }]
+    if dist.is_initialized():
+        dist.broadcast_object_list(payload, src=0)
+    return payload[0]
+
+
+def run_1f1b_training(cfg: BenchConfig) -> dict:
+    set_default_backend(cfg.kernel_backend)
+    env = init_distributed()
+    world = env.world_size
+    device = resolve_device(cfg, env.local_rank)
+    dtype = compute_dtype(cfg)
+
+    torch.manual_seed(cfg.seed)  # identical init everywhere
+    seq = build_sequential(cfg.dataset, cfg.arch)
+    if dtype != torch.float32:
+        seq = seq.to(dtype)
+
+    plan_info = _make_plan(cfg, seq, device, world)
+    m2s = plan_info["module_to_stage_map"]
+    plan = StagePlan(replicas=plan_info["replicas"])
+    assert plan.world_size == world, (plan.replicas, world)
+
+    stage, replica = plan.stage_of_rank(env.rank)
+    my_layers = [i for i, s in enumerate(m2s) if s == stage]
+    stage_mod = torch.nn.Sequential(
+        *[seq[i] for i in my_layers]).to(device)
+
+    # static shapes from a tiny CPU dry run, batch dim patched to B
+    stage_slices = []
+    off = 0
+    for s in range(plan.num_stages):
+        layers = [i for i, t in enumerate(m2s) if t == s]
+        stage_slices.append(torch.nn.Sequential(*[seq[i] for i in layers]))
+    probe = torch.zeros((2,) + tuple(cfg.shape), dtype=dtype)
+    shapes = dry_run_shapes(stage_slices, probe)
+    B = cfg.batch_size
+
+    def with_batch(shape):
+        return torch.Size((B,) + tuple(shape)[1:])
+
+    in_shape = (None if stage == 0
+                else with_batch(shapes[stage - 1]))
+    out_shape = with_batch(shapes[stage])
+
+    backend = dist.get_backend() if dist.is_initialized() else "gloo"
+    transport = PipelineTransport(plan.edges(), backend)
+
+    # per-stage DP groups (every rank participates in every new_group)
+    dp = None
+    for s in range(plan.num_stages):
+        ranks = plan.stage_ranks(s)
+        if len(ranks) > 1:
+            g = dist.new_group(ranks)
+            if s == stage:
+                dp = BucketedDataParallel(stage_mod, process_group=g)
+
+    loss_fn = (lambda out, tgt: NF.cross_entropy(
+        out, tgt, backend=cfg.kernel_backend))
+    rt = StageRuntime(plan, env.rank, stage_mod, transport, in_shape,
+                      out_shape, device, dtype, loss_fn=loss_fn,
+                      dp_wrapper=dp)
+
+    opt = VersionedOptimizer(
+        FusedSGD(stage_mod.parameters(), lr=cfg.lr, momentum=cfg.momentum,
+                 weight_decay=cfg.weight_decay,
+                 backend=cfg.kernel_backend),
+        versioned=plan.num_warmup(stage) > 0)
+
+    # deterministic synthetic streams: stage 0 reads inputs, last stage
+    # reads the matching targets (no label transport needed)
+    train_ds = SyntheticImageDataset(cfg.dataset, train=True,
+                                     seed=cfg.seed,
+                                     scale=cfg.synthetic_scale)
+    test_ds = SyntheticImageDataset(cfg.dataset, train=False,
+                                    seed=cfg.seed,
+                                    scale=cfg.synthetic_scale)
+
+    def make_providers(ds, epoch):
+        n_mb = len(ds) // B
+        lcm = plan.lcm_replicas()
+        n_mb = max((n_mb // lcm) * lcm, 0)
+        g = torch.Generator().manual_seed(cfg.seed * 977 + epoch)
+        perm = torch.randperm(len(ds), generator=g)
+
+        def input_provider(mb):
+            idx = perm[mb * B:(mb + 1) * B]
+            return torch.stack([ds[i.item()][0] for i in idx])
+
+        def target_provider(mb):
+            idx = perm[mb * B:(mb + 1) * B]
+            return torch.tensor([ds[i.item()][1] for i in idx])
+
+        return n_mb, input_provider, target_provider
+
+    is_output = rt.is_last and replica == 0
+    log = BenchLogger(0 if is_output else 1)
+
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    def train_epoch(epoch):
+        stage_mod.train()
+        n_mb, inp, tgt = make_providers(train_ds, epoch)
+        mbs = rt.my_minibatches(n_mb)
+        warmup = min(plan.num_warmup(stage), len(mbs))
+        loss_vals = []  # detached tensors; .item() deferred to epoch end
+        if dist.is_initialized():
+            dist.barrier()
+        sync()
+        t0 = time.perf_counter()
+        for k in range(warmup):
+            loss, _ = rt.run_forward(mbs[k], inp, tgt, training=True)
+            if loss is not None:
+                loss_vals.append(loss.detach())
+        for k in range(len(mbs)):
+            if warmup + k < len(mbs):
+                loss, _ = rt.run_forward(mbs[warmup + k], inp, tgt,
+                                         training=True)
+                if loss is not None:
+                    loss_vals.append(loss.detach())
+            if dp is not None:
+                dp.zero_grad_buckets()
+            else:
+                opt.zero_grad(set_to_none=False)
+            rt.run_backward()
+            if dp is not None:
+                dp.finalize_backward()
+            opt.step()
+        sync()
+        if dist.is_initialized():
+            dist.barrier()
+        elapsed = time.perf_counter() - t0
+        sps = n_mb * B / max(elapsed, 1e-9)
+        avg_loss = (torch.stack(loss_vals).mean().item()
+                    if loss_vals else 0.0)
+        return avg_loss, sps, elapsed
+
+    @torch.no_grad()
+    def validate():
+        stage_mod.eval()
+        n_mb, inp, tgt = make_providers(test_ds, 0)
+        mbs = rt.my_minibatches(n_mb)
+        window = plan.num_stages
+        losses = AverageMeter()
+        accs = AverageMeter()
+        outstanding = 0
+        for k, mb in enumerate(mbs):
+            loss, extras = rt.run_forward(mb, inp, tgt, training=False)
+            # per-hop credit: ack upstream as soon as mb is consumed
+            if not rt.is_first:
+                rt.send_ack(mb)
+            if rt.is_last:
+                y = extras["target"]
+                out = extras["output"]
+                losses.update(loss.item(), y.numel())
+                accs.update(accuracy(out, y)[0], y.numel())
+                rt.pop_eval()
+            else:
+                outstanding += 1
+                if outstanding >= window:
+                    mb_old, _ = rt.pop_eval()
+                    rt.recv_ack(mb_old)
+                    outstanding -= 1
+        while outstanding > 0 and not rt.is_last:
+            mb_old, _ = rt.pop_eval()
+            rt.recv_ack(mb_old)
+            outstanding -= 1
+        if dist.is_initialized():
+            dist.barrier()
+        return losses.avg, accs.avg
+
+    epoch_sps, epoch_secs = [], []
+    val_loss = val_acc = 0.0
+    train_loss = 0.0
+    for epoch in range(1, cfg.epochs + 1):
+        train_loss, sps, secs = train_epoch(epoch)
+        val_loss, val_acc = validate()
+        epoch_sps.append(sps)
+        epoch_secs.append(secs)
+        log.epoch(epoch, cfg.epochs, train_loss, sps, val_loss, val_acc)
+    avg_sps = sum(epoch_sps) / len(epoch_sps)
+    avg_secs = sum(epoch_secs) / len(epoch_secs)
+    log.final(val_acc, avg_sps, avg_secs)
+    return {"valid_accuracy": val_acc, "samples_per_sec": avg_sps,
+            "sec_per_epoch": avg_secs, "stage": stage,
+            "replicas": plan.replicas}

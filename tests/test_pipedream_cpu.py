@@ -1,0 +1,125 @@
+"""1F1B pipeline runtime on gloo, world_size 2, CPU.
+
+Covers: p2p transport, model-parallel exact-match vs single process,
+and the full profile->partition->1F1B runner smoke (the multi-process
+test pyramid the reference lacks — SURVEY.md §4)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+
+def _env(rank, world, port):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+
+
+# --------------------------------------------------------------- transport
+def _worker_transport(rank, world, port):
+    _env(rank, world, port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
+    tr = PipelineTransport([(0, 1)], backend="gloo")
+    if rank == 0:
+        t = torch.arange(12, dtype=torch.float32).reshape(3, 4)
+        tr.channel(0, 1, "fwd").isend(t).wait()
+        buf = torch.empty(3, 4)
+        tr.channel(0, 1, "bwd").irecv(buf).wait()
+        assert torch.equal(buf, t * 2)
+    else:
+        buf = torch.empty(3, 4)
+        tr.channel(0, 1, "fwd").irecv(buf).wait()
+        tr.channel(0, 1, "bwd").isend(buf * 2).wait()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_transport_roundtrip(free_port):
+    mp.spawn(_worker_transport, args=(2, free_port), nprocs=2, join=True)
+
+
+# ------------------------------------------- model parallel == sequential
+def _worker_mp_exact(rank, world, port):
+    _env(rank, world, port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from ddlbench_amd.ops.sgd import FusedSGD
+    from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
+    from ddlbench_amd.parallel.pipeline.runtime import (StagePlan,
+                                                        StageRuntime)
+
+    torch.manual_seed(0)
+    full = torch.nn.Sequential(
+        torch.nn.Linear(6, 16), torch.nn.Tanh(),
+        torch.nn.Linear(16, 8), torch.nn.Tanh(), torch.nn.Linear(8, 3))
+    stage_mods = [torch.nn.Sequential(*list(full)[:2]),
+                  torch.nn.Sequential(*list(full)[2:])]
+    plan = StagePlan(replicas=[1, 1])
+    tr = PipelineTransport(plan.edges(), backend="gloo")
+    mod = stage_mods[rank]
+    dev = torch.device("cpu")
+    loss_fn = torch.nn.functional.cross_entropy
+    B = 4
+    rt = StageRuntime(plan, rank, mod, tr,
+                      in_shape=torch.Size([B, 16]) if rank else None,
+                      out_shape=torch.Size([B, 16]) if rank == 0
+                      else torch.Size([B, 3]),
+                      device=dev, dtype=torch.float32, loss_fn=loss_fn)
+    opt = FusedSGD(mod.parameters(), lr=0.1, momentum=0.9, backend="torch")
+
+    gen = torch.Generator().manual_seed(7)
+    xs = [torch.randn(B, 6, generator=gen) for _ in range(5)]
+    ys = [torch.randint(3, (B,), generator=gen) for _ in range(5)]
+
+    # no pipelining (warmup=0): fwd+bwd+step per minibatch == sequential
+    for m in range(5):
+        rt.run_forward(m, lambda i: xs[i], lambda i: ys[i], training=True)
+        opt.zero_grad(set_to_none=False)
+        rt.run_backward()
+        opt.step()
+
+    # single-process reference
+    torch.manual_seed(0)
+    ref = torch.nn.Sequential(
+        torch.nn.Linear(6, 16), torch.nn.Tanh(),
+        torch.nn.Linear(16, 8), torch.nn.Tanh(), torch.nn.Linear(8, 3))
+    ropt = FusedSGD(ref.parameters(), lr=0.1, momentum=0.9, backend="torch")
+    for m in range(5):
+        ropt.zero_grad(set_to_none=False)
+        loss_fn(ref(xs[m]), ys[m]).backward()
+        ropt.step()
+
+    ref_stage = [torch.nn.Sequential(*list(ref)[:2]),
+                 torch.nn.Sequential(*list(ref)[2:])][rank]
+    for p, q in zip(mod.parameters(), ref_stage.parameters()):
+        torch.testing.assert_close(p, q, rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_model_parallel_matches_sequential(free_port):
+    mp.spawn(_worker_mp_exact, args=(2, free_port), nprocs=2, join=True)
+
+
+# ------------------------------------------------------ full runner smoke
+def _worker_runner(rank, world, port):
+    _env(rank, world, port)
+    from ddlbench_amd.config import BenchConfig
+    from ddlbench_amd.parallel.pipeline.runner import run_1f1b_training
+    cfg = BenchConfig(dataset="mnist", arch="resnet18",
+                      strategy="pipedream", epochs=1, batch_size=8,
+                      synthetic_scale=0.0008, device="cpu",
+                      num_workers=0, log_interval=0)
+    res = run_1f1b_training(cfg)
+    assert res["samples_per_sec"] > 0
+    assert torch.isfinite(torch.tensor(res["valid_accuracy"]))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_full_1f1b_runner_two_stages(free_port):
+    mp.spawn(_worker_runner, args=(2, free_port), nprocs=2, join=True)
