@@ -66,6 +66,8 @@ class BenchConfig:
     seed: int = 42
     num_workers: int = 2
     kernel_backend: str = "auto"      # auto | native | torch (ops dispatch)
+    checkpoint_dir: str = ""          # per-stage checkpoints when set
+    resume: bool = False              # load stage checkpoints at start
 
     def __post_init__(self) -> None:
         if self.dataset not in DATASET_SHAPES:

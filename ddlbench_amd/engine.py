@@ -134,11 +134,12 @@ class Trainer:
         return (self.allreduce_metrics(losses.avg),
                 self.allreduce_metrics(accs.avg))
 
-    def fit(self, train_loader, test_loader, train_sampler=None) -> dict:
+    def fit(self, train_loader, test_loader, train_sampler=None,
+            start_epoch: int = 1, on_epoch_end=None) -> dict:
         cfg = self.cfg
         epoch_sps, epoch_secs = [], []
         val_acc = val_loss = 0.0
-        for epoch in range(1, cfg.epochs + 1):
+        for epoch in range(start_epoch, cfg.epochs + 1):
             if train_sampler is not None:
                 train_sampler.set_epoch(epoch)
             train_loss, sps, secs = self.train_epoch(train_loader, epoch)
@@ -147,8 +148,11 @@ class Trainer:
             epoch_secs.append(secs)
             self.log.epoch(epoch, cfg.epochs, train_loss, sps,
                            val_loss, val_acc)
-        avg_sps = sum(epoch_sps) / len(epoch_sps)
-        avg_secs = sum(epoch_secs) / len(epoch_secs)
+            if on_epoch_end is not None:
+                on_epoch_end(epoch, {"valid_accuracy": val_acc,
+                                     "valid_loss": val_loss})
+        avg_sps = sum(epoch_sps) / max(len(epoch_sps), 1)
+        avg_secs = sum(epoch_secs) / max(len(epoch_secs), 1)
         self.log.final(val_acc, avg_sps, avg_secs)
         return {"valid_accuracy": val_acc, "samples_per_sec": avg_sps,
                 "sec_per_epoch": avg_secs}

@@ -233,17 +233,31 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
             dist.barrier()
         return losses.avg, accs.avg
 
+    start_epoch = 1
+    if cfg.resume and cfg.checkpoint_dir:
+        from ddlbench_amd.utils.checkpoint import load_stage_checkpoint
+        state = load_stage_checkpoint(cfg.checkpoint_dir, stage, stage_mod,
+                                      opt, map_location=device)
+        if state is not None:
+            start_epoch = state["epoch"] + 1
+
     epoch_sps, epoch_secs = [], []
     val_loss = val_acc = 0.0
     train_loss = 0.0
-    for epoch in range(1, cfg.epochs + 1):
+    for epoch in range(start_epoch, cfg.epochs + 1):
         train_loss, sps, secs = train_epoch(epoch)
         val_loss, val_acc = validate()
         epoch_sps.append(sps)
         epoch_secs.append(secs)
         log.epoch(epoch, cfg.epochs, train_loss, sps, val_loss, val_acc)
-    avg_sps = sum(epoch_sps) / len(epoch_sps)
-    avg_secs = sum(epoch_secs) / len(epoch_secs)
+        # per-stage checkpoint by rank_in_stage 0 (reference
+        # main_with_runtime.py:393-403)
+        if cfg.checkpoint_dir and replica == 0:
+            from ddlbench_amd.utils.checkpoint import save_stage_checkpoint
+            save_stage_checkpoint(cfg.checkpoint_dir, stage, epoch,
+                                  cfg.arch, stage_mod, opt, val_acc)
+    avg_sps = sum(epoch_sps) / max(len(epoch_sps), 1)
+    avg_secs = sum(epoch_secs) / max(len(epoch_secs), 1)
     log.final(val_acc, avg_sps, avg_secs)
     return {"valid_accuracy": val_acc, "samples_per_sec": avg_sps,
             "sec_per_epoch": avg_secs, "stage": stage,

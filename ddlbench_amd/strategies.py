@@ -32,10 +32,26 @@ def run_single(cfg: BenchConfig) -> dict:
     if compute_dtype(cfg) != torch.float32:
         model = model.to(compute_dtype(cfg))
     optimizer = make_optimizer(cfg, model)
+    start_epoch = 1
+    if cfg.resume and cfg.checkpoint_dir:
+        from ddlbench_amd.utils.checkpoint import load_stage_checkpoint
+        state = load_stage_checkpoint(cfg.checkpoint_dir, 0, model,
+                                      optimizer)
+        if state is not None:
+            start_epoch = state["epoch"] + 1
     train_loader, test_loader, _ = make_loaders(
         cfg, pin_memory=device.type == "cuda")
     trainer = Trainer(cfg, model, optimizer, device)
-    return trainer.fit(train_loader, test_loader)
+
+    def on_epoch_end(epoch, metrics):
+        if cfg.checkpoint_dir:
+            from ddlbench_amd.utils.checkpoint import save_stage_checkpoint
+            save_stage_checkpoint(cfg.checkpoint_dir, 0, epoch, cfg.arch,
+                                  model, optimizer,
+                                  metrics.get("valid_accuracy", 0.0))
+
+    return trainer.fit(train_loader, test_loader,
+                       start_epoch=start_epoch, on_epoch_end=on_epoch_end)
 
 
 def run_ddp(cfg: BenchConfig) -> dict:
