@@ -42,6 +42,9 @@ def parse_args():
                    help="auto|cuda|cpu (cpu only for plumbing tests)")
     p.add_argument("--kernel-backend", default="auto",
                    choices=["auto", "native", "torch"])
+    p.add_argument("--conv", default="miopen", choices=["miopen", "mfma"],
+                   help="conv backend: library (MIOpen) or the in-tree "
+                        "MFMA implicit-GEMM kernels")
     p.add_argument("--memory-format", default="channels_last",
                    choices=["channels_last", "contiguous"],
                    help="channels_last (NHWC) keeps MIOpen on its native "
@@ -80,6 +83,12 @@ def main():
         model = model.to(dtype)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
+    if args.conv == "mfma" and device.type == "cuda":
+        from ddlbench_amd.ops.conv import convert_convs
+        n_conv = convert_convs(model, dtype)
+        if env.rank == 0:
+            print(f"# mfma conv kernels on {n_conv} layers",
+                  file=sys.stderr)
     dp = BucketedDataParallel(model)
     opt = FusedSGD(model.parameters(), lr=0.1 * world, momentum=0.9,
                    weight_decay=1e-4, backend=cfg.kernel_backend)

@@ -43,6 +43,10 @@ template <typename T>
 void launch_ce_bwd(const T*, const int64_t*, const float*, const float*, T*,
                    int64_t, int64_t, hipStream_t);
 
+void launch_conv_igemm(const void*, const void*, void*, const void*, int,
+                       int, int, int, int, int, int, int, int, int, int,
+                       int, hipStream_t);
+
 template <typename T>
 void launch_dw3x3_fwd(const T*, const float*, T*, int64_t, int64_t, int64_t,
                       int64_t, int64_t, int64_t, int, hipStream_t);
@@ -237,6 +241,57 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor target,
   return dx;
 }
 
+// ---- MFMA implicit-GEMM conv (NHWC bf16) ------------------------------
+// x: (N,C,H,W) logical, channels_last memory; w: (K,C,R,S) logical,
+// channels_last memory (= (K,R,S,C) image). Returns channels_last y.
+torch::Tensor conv_igemm_fwd(torch::Tensor x, torch::Tensor w,
+                             int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "conv_igemm: bf16 HIP tensors only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels_last");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "w must be channels_last");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(C % 8 == 0, "conv_igemm fwd needs C % 8 == 0");
+  const int OH = (H + 2 * (int)pad - R) / (int)stride + 1;
+  const int OW = (W + 2 * (int)pad - S) / (int)stride + 1;
+  auto y = torch::empty({N, K, OH, OW},
+                        x.options().memory_format(
+                            at::MemoryFormat::ChannelsLast));
+  auto zero = torch::zeros({16}, x.options());
+  launch_conv_igemm(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                    zero.data_ptr(), N, H, W, C, K, OH, OW, R, S,
+                    (int)stride, (int)pad, /*dgrad=*/0,
+                    cur_stream());
+  return y;
+}
+
+// dy: (N,K,OH,OW) channels_last; w_perm: (C,R,S,K) plain-contiguous
+// memory (host permutes once per backward). Returns channels_last dx.
+torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
+                               int64_t N, int64_t C, int64_t H, int64_t W,
+                               int64_t stride, int64_t pad) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16,
+              "conv_igemm: bf16 HIP tensors only");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "dy must be channels_last");
+  TORCH_CHECK(w_perm.is_contiguous(), "w_perm must be contiguous");
+  const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  const int R = w_perm.size(1), S = w_perm.size(2);
+  TORCH_CHECK(K % 8 == 0, "conv_igemm dgrad needs K % 8 == 0");
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(
+                             at::MemoryFormat::ChannelsLast));
+  auto zero = torch::zeros({16}, dy.options());
+  launch_conv_igemm(dy.data_ptr(), w_perm.data_ptr(), dx.data_ptr(),
+                    zero.data_ptr(), (int)N, (int)H, (int)W, (int)C, K,
+                    OH, OW, R, S, (int)stride, (int)pad, /*dgrad=*/1,
+                    cur_stream());
+  return dx;
+}
+
 // ---- depthwise 3x3 ----------------------------------------------------
 torch::Tensor dw3x3_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
   check_gpu_contig(x, "x");
@@ -294,6 +349,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD step");
   m.def("bn_act_fwd", &bn_act_fwd, "fused BN+act(+res) forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act(+res) backward");
+  m.def("conv_igemm_fwd", &conv_igemm_fwd,
+        "NHWC bf16 MFMA implicit-GEMM conv forward");
+  m.def("conv_igemm_dgrad", &conv_igemm_dgrad,
+        "NHWC bf16 MFMA implicit-GEMM conv data-grad");
   m.def("ce_fwd", &ce_fwd, "cross-entropy forward");
   m.def("ce_bwd", &ce_bwd, "cross-entropy backward");
   m.def("dw3x3_fwd", &dw3x3_fwd, "depthwise 3x3 forward");

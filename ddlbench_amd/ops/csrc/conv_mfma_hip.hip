@@ -1,0 +1,273 @@
+#include "hip/hip_runtime.h"
+// NHWC implicit-GEMM convolution on MFMA matrix cores (gfx950, bf16).
+//
+// The conv/GEMM core the reference gets from cuDNN (SURVEY.md §2.11
+// "MI355X plan implication"), hand-written for CDNA4:
+//   y[n,oh,ow,k] = sum_{r,s,c} x[n, oh*st+r-p, ow*st+s-p, c] * w[k,r,s,c]
+// viewed as GEMM  C[M=N*OH*OW][Nd=K] = A[M][Kd=R*S*C] * B[Nd][Kd]^T
+// with A materialized implicitly (im2col addressing in the staging
+// stage) and B = the conv weight read directly in its channels-last
+// (K,R,S,C) memory image — no host-side im2col, no weight transform for
+// forward.
+//
+// One kernel template serves forward (FPROP) and data-grad (DGRAD —
+// A = scatter-gathered dy with transposed-conv addressing, B = the
+// weight permuted to (C,R,S,K) memory once per backward).
+//
+// Structure = the documented CDNA4 GEMM recipe (cdna_hip_programming.md
+// §5): 128x128 block tile, BK=64, 4 waves each owning a 64x64 sub-tile
+// of 16x16x32 bf16 MFMA fragments, double-buffered LDS filled by
+// 16-byte global_load_lds (lane-linear dest; XOR bank swizzle applied on
+// the SOURCE chunk index and re-applied on the ds_read side — rule 21),
+// out-of-bounds/padding chunks redirected to a zero page so every lane
+// always issues its DMA.
+//
+// Constraints (dispatcher falls back for the rest): bf16, groups=1,
+// dilation=1, C %8 == 0 (FPROP) / K %8 == 0 (DGRAD) so no 16-B chunk
+// crosses an (r,s) boundary.
+
+#include "common.h"
+#include <stdint.h>
+#include <stdexcept>
+#include <string>
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define THREADS 256
+
+struct ConvParams {
+  const bf16* a;      // x (FPROP) or dy (DGRAD), NHWC memory
+  const bf16* b;      // w as (K,R,S,C) mem (FPROP) or (C,R,S,K) (DGRAD)
+  bf16* out;          // y (FPROP: M x K) or dx (DGRAD: M x C)
+  const bf16* zero;   // >=16B of zeros
+  int N, H, W, Cin;   // logical input dims of the *forward* conv
+  int K, OH, OW;      // output channels / spatial of the forward conv
+  int R, S, stride, pad;
+  long M, Nd, Kd;     // GEMM dims of THIS pass
+};
+
+// ---- chunk -> global address generators -------------------------------
+// A-chunk: logical (row=m in [0,BM), cg in [0,8)) of the current K-step.
+// Returns the 16-byte-aligned source for elements kk0+cg*8 .. +7.
+
+template <bool DGRAD>
+DEV const bf16* a_chunk_addr(const ConvParams& p, long m, long kkg) {
+  if (m >= p.M) return p.zero;
+  if (!DGRAD) {
+    // m -> (n, oh, ow); kkg -> (r, s, c0)
+    const int ohw = p.OH * p.OW;
+    const int n = (int)(m / ohw);
+    const int rem = (int)(m - (long)n * ohw);
+    const int oh = rem / p.OW, ow = rem - (rem / p.OW) * p.OW;
+    const int c0 = (int)(kkg % p.Cin);
+    const int rs = (int)(kkg / p.Cin);
+    const int r = rs / p.S, s = rs - (rs / p.S) * p.S;
+    const int ih = oh * p.stride + r - p.pad;
+    const int iw = ow * p.stride + s - p.pad;
+    if (ih < 0 || ih >= p.H || iw < 0 || iw >= p.W) return p.zero;
+    return p.a + (((long)n * p.H + ih) * p.W + iw) * p.Cin + c0;
+  } else {
+    // m -> (n, ih, iw); kkg -> (r, s, k0); gather from dy
+    const int hw = p.H * p.W;
+    const int n = (int)(m / hw);
+    const int rem = (int)(m - (long)n * hw);
+    const int ih = rem / p.W, iw = rem - (rem / p.W) * p.W;
+    const int k0 = (int)(kkg % p.K);
+    const int rs = (int)(kkg / p.K);
+    const int r = rs / p.S, s = rs - (rs / p.S) * p.S;
+    const int tih = ih + p.pad - r;
+    const int tiw = iw + p.pad - s;
+    if (tih < 0 || tiw < 0 || (tih % p.stride) || (tiw % p.stride))
+      return p.zero;
+    const int oh = tih / p.stride, ow = tiw / p.stride;
+    if (oh >= p.OH || ow >= p.OW) return p.zero;
+    return p.a + (((long)n * p.OH + oh) * p.OW + ow) * p.K + k0;
+  }
+}
+
+DEV const bf16* b_chunk_addr(const ConvParams& p, long row, long kkg) {
+  if (row >= p.Nd || kkg >= p.Kd) return p.zero;
+  return p.b + row * p.Kd + kkg;
+}
+
+// ---- the kernel -------------------------------------------------------
+// LDS: A[2][128][64] + B[2][128][64] bf16 = 64 KiB. Lane-linear glds
+// image: slot ca in [0,1024) holds logical (row = ca>>3, cg' = ca&7)
+// where the DATA stored is logical cg = cg' ^ (row & 7). ds_read applies
+// the same XOR.
+
+template <bool DGRAD>
+__global__ __launch_bounds__(THREADS, 2)
+void conv_igemm_kernel(ConvParams p) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* lds = reinterpret_cast<bf16*>(smem);
+  // buffers: [buf][AB][128*64]
+  auto lds_tile = [&](int buf, int ab) {
+    return lds + ((buf * 2 + ab) * BM * BK);
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  const int nbn = (int)((p.Nd + BN - 1) / BN);
+  int block = blockIdx.x;
+  // XCD-aware swizzle: contiguous chunks per XCD (bijective form)
+  {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rmd = nwg % 8;
+    const int xcd = block % 8, idx = block / 8;
+    block = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
+            + idx;
+  }
+  const long bm = (long)(block / nbn) * BM;
+  const long bn = (long)(block % nbn) * BN;
+
+  const int nsteps = (int)((p.Kd + BK - 1) / BK);
+
+  // ---- staging: each thread owns 4 A-chunks and 4 B-chunks ----------
+  // slot ca = l*256 + tid; row = ca>>3, cg_store = ca&7,
+  // data(cg) = cg_store ^ (row&7)
+  int a_row[4], a_cg[4], b_row[4], b_cg[4];
+#pragma unroll
+  for (int l = 0; l < 4; ++l) {
+    const int ca = l * THREADS + tid;
+    a_row[l] = ca >> 3;
+    a_cg[l] = (ca & 7) ^ (a_row[l] & 7);
+    b_row[l] = ca >> 3;
+    b_cg[l] = (ca & 7) ^ (b_row[l] & 7);
+  }
+
+  auto stage = [&](int buf, int step) {
+    const long kk0 = (long)step * BK;
+    bf16* la = lds_tile(buf, 0);
+    bf16* lb = lds_tile(buf, 1);
+#pragma unroll
+    for (int l = 0; l < 4; ++l) {
+      const long kkga = kk0 + a_cg[l] * 8;
+      const bf16* src = a_chunk_addr<DGRAD>(p, bm + a_row[l],
+                                            kkga < p.Kd ? kkga : 0);
+      if (kkga >= p.Kd) src = p.zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(la +
+              (l * THREADS + tid) * 8), 16, 0, 0);
+    }
+#pragma unroll
+    for (int l = 0; l < 4; ++l) {
+      const bf16* src = b_chunk_addr(p, bn + b_row[l],
+                                     kk0 + b_cg[l] * 8);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lb +
+              (l * THREADS + tid) * 8), 16, 0, 0);
+    }
+  };
+
+  // ---- fragment read offsets (XOR re-applied) -----------------------
+  // A frag (16x16x32): lane reads row m0 + (lane&15), elems
+  // ks*32 + (lane>>4)*8 .. +7  ->  cg = ks*4 + (lane>>4)
+  const int fr = lane & 15;         // row-in-frag
+  const int fk = lane >> 4;         // k-subgroup
+  const int wm = (wid >> 1) * 64;   // wave M offset
+  const int wn = (wid & 1) * 64;    // wave N offset
+
+  auto frag_ptr = [&](bf16* tile, int row, int ks) -> const bf16x8* {
+    const int cg = (ks * 4 + fk) ^ (row & 7);
+    return reinterpret_cast<const bf16x8*>(tile + row * BK + cg * 8);
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = 0; t < nsteps; ++t) {
+    if (t + 1 < nsteps) stage(cur ^ 1, t + 1);
+    bf16* la = lds_tile(cur, 0);
+    bf16* lb = lds_tile(cur, 1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        af[mf] = *frag_ptr(la, wm + mf * 16 + fr, ks);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        bfr[nf] = *frag_ptr(lb, wn + nf * 16 + fr, ks);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: D[row=(lane>>4)*4+reg][col=lane&15] per fragment ---
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const long row = bm + wm + mf * 16 + fk * 4 + reg;
+      if (row >= p.M) continue;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const long col = bn + wn + nf * 16 + fr;
+        if (col < p.Nd)
+          p.out[row * p.Nd + col] = from_f32<bf16>(acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
+// ---- launchers --------------------------------------------------------
+void launch_conv_igemm(const void* a, const void* b, void* out,
+                       const void* zero, int N, int H, int W, int Cin,
+                       int K, int OH, int OW, int R, int S, int stride,
+                       int pad, int dgrad, hipStream_t stream) {
+  ConvParams p;
+  p.a = (const bf16*)a;
+  p.b = (const bf16*)b;
+  p.out = (bf16*)out;
+  p.zero = (const bf16*)zero;
+  p.N = N; p.H = H; p.W = W; p.Cin = Cin; p.K = K; p.OH = OH; p.OW = OW;
+  p.R = R; p.S = S; p.stride = stride; p.pad = pad;
+  if (!dgrad) {
+    p.M = (long)N * OH * OW;
+    p.Nd = K;
+    p.Kd = (long)R * S * Cin;
+  } else {
+    p.M = (long)N * H * W;
+    p.Nd = Cin;
+    p.Kd = (long)R * S * K;
+  }
+  const long nbm = (p.M + BM - 1) / BM;
+  const long nbn = (p.Nd + BN - 1) / BN;
+  const size_t lds_bytes = 2 * 2 * BM * BK * sizeof(bf16);
+  const dim3 grid((unsigned)(nbm * nbn));
+  if (!dgrad)
+    hipLaunchKernelGGL((conv_igemm_kernel<false>), grid, dim3(THREADS),
+                       lds_bytes, stream, p);
+  else
+    hipLaunchKernelGGL((conv_igemm_kernel<true>), grid, dim3(THREADS),
+                       lds_bytes, stream, p);
+  HIP_CHECK_LAST();
+}

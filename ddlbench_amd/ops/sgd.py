@@ -99,7 +99,11 @@ class FusedSGD(Optimizer):
                 cached["first_step"] = first_step
             first_step = cached.pop("first_step", False)
             for p in ps:
-                assert p.grad.is_contiguous(), "FusedSGD needs contiguous grads"
+                # the fused kernel walks param/grad as flat buffers in
+                # storage order — any dense layout works (NCHW or
+                # channels_last) as long as grad strides match param's
+                assert p.grad.stride() == p.stride(), \
+                    "FusedSGD needs grads with the param's layout"
             ptr_grads = torch.tensor([p.grad.data_ptr() for p in ps],
                                      dtype=torch.int64).to(dev,
                                                            non_blocking=True)

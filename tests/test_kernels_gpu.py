@@ -76,7 +76,10 @@ def test_bn_act_forward_backward(dtype, tol, act, with_res):
     if act != "none" and dtype is torch.bfloat16:
         band = (y2 - 0.0).abs() > 0.05
         if act == "relu6":
-            band &= (y2 - 6.0).abs() > 0.05
+            # bf16 ulp near 6.0 is 0.03125 and the bf16-input BN output
+            # can differ from the fp32 reference by a few 1e-2 — any
+            # element within 0.3 of the clamp can land on either side
+            band &= (y2 - 6.0).abs() > 0.3
     torch.testing.assert_close(x1.grad.float()[band], x2.grad[band],
                                rtol=tol, atol=tol * 10)
     torch.testing.assert_close(g1.grad, g2.grad, rtol=1e-3, atol=1e-3)
@@ -186,6 +189,62 @@ def test_depthwise_conv(dtype, tol, stride):
     torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=tol, atol=tol)
     torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=tol,
                                atol=tol * 10)
+
+
+# ----------------------------------------------- MFMA implicit-GEMM conv
+@pytest.mark.parametrize("shape", [
+    # (N, C, H, W, K, R, stride, pad)
+    (4, 64, 14, 14, 128, 1, 1, 0),      # 1x1
+    (4, 64, 14, 14, 64, 3, 1, 1),       # 3x3 s1
+    (4, 64, 15, 15, 128, 3, 2, 1),      # 3x3 s2, odd spatial
+    (2, 128, 7, 7, 120, 3, 1, 1),       # Nd not multiple of tile
+    (2, 16, 9, 9, 24, 3, 2, 1),         # small C/K (mobilenet-ish)
+])
+def test_conv_mfma_fwd_dgrad(shape):
+    from ddlbench_amd.ops.conv import conv2d_mfma
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(0)
+    dev = _dev()
+    x1 = torch.randn(N, C, H, W, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    w1 = torch.randn(K, C, R, R, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y1 = conv2d_mfma(x1, w1, stride, pad)
+    dy = torch.randn_like(y1)
+    y1.backward(dy)
+
+    x2 = x1.detach().float().contiguous().requires_grad_(True)
+    w2 = w1.detach().float().contiguous().requires_grad_(True)
+    y2 = torch.nn.functional.conv2d(x2, w2, None, stride, pad)
+    y2.backward(dy.float().contiguous())
+
+    kd = R * R * C
+    tol = 0.03 * (kd ** 0.5) / 8  # bf16 accum noise grows with sqrt(K)
+    torch.testing.assert_close(y1.float().contiguous(), y2,
+                               rtol=5e-2, atol=max(tol, 0.05))
+    torch.testing.assert_close(x1.grad.float().contiguous(), x2.grad,
+                               rtol=5e-2, atol=max(tol, 0.05))
+    torch.testing.assert_close(w1.grad.float().contiguous(), w2.grad,
+                               rtol=5e-2, atol=1.0)
+
+
+def test_conv_mfma_resnet_block_trains():
+    """convert_convs on a resnet block; one train step, finite loss."""
+    from ddlbench_amd.models import build_model
+    from ddlbench_amd.ops.conv import convert_convs
+    torch.manual_seed(0)
+    dev = _dev()
+    m = build_model("cifar10", "resnet18").to(dev).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    n = convert_convs(m)
+    assert n > 10
+    x = torch.randn(8, 3, 32, 32, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(10, (8,), device=dev)
+    out = m(x)
+    loss = torch.nn.functional.cross_entropy(out.float(), y)
+    loss.backward()
+    assert torch.isfinite(loss)
 
 
 # ------------------------------------------------------------ fused SGD
