@@ -48,6 +48,12 @@ void launch_conv_igemm(const void*, const void*, void*, const void*, int,
                        int, hipStream_t);
 
 template <typename T>
+void launch_revert_varlen(const T*, T*, const int64_t*, int64_t, int64_t,
+                          int64_t, hipStream_t);
+void launch_varlen_mask(const int64_t*, uint8_t*, int64_t, int64_t,
+                        hipStream_t);
+
+template <typename T>
 void launch_dw3x3_fwd(const T*, const float*, T*, int64_t, int64_t, int64_t,
                       int64_t, int64_t, int64_t, int, hipStream_t);
 template <typename T>
@@ -344,6 +350,33 @@ std::vector<torch::Tensor> dw3x3_bwd(torch::Tensor x, torch::Tensor w,
   return {dx, dwt};
 }
 
+// ---- sequence utils (GNMT) --------------------------------------------
+torch::Tensor revert_varlen(torch::Tensor x, torch::Tensor lengths) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.dim() == 3, "x must be (T, B, F)");
+  TORCH_CHECK(lengths.scalar_type() == at::kLong, "lengths must be int64");
+  const int64_t T = x.size(0), B = x.size(1), F = x.size(2);
+  auto out = torch::empty_like(x);
+  auto s = cur_stream();
+  if (is_bf16(x))
+    launch_revert_varlen<__hip_bfloat16>(
+        dptr<__hip_bfloat16>(x), dptr<__hip_bfloat16>(out),
+        dptr<int64_t>(lengths), T, B, F, s);
+  else
+    launch_revert_varlen<float>(dptr<float>(x), dptr<float>(out),
+                                dptr<int64_t>(lengths), T, B, F, s);
+  return out;
+}
+
+torch::Tensor varlen_mask(torch::Tensor lengths, int64_t T) {
+  TORCH_CHECK(lengths.is_cuda(), "lengths must be on the HIP device");
+  const int64_t B = lengths.size(0);
+  auto mask = torch::empty({T, B}, lengths.options().dtype(at::kByte));
+  launch_varlen_mask(dptr<int64_t>(lengths), dptr<uint8_t>(mask), T, B,
+                     cur_stream());
+  return mask;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "ddlbench_amd gfx950 HIP kernels";
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD step");
@@ -355,6 +388,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC bf16 MFMA implicit-GEMM conv data-grad");
   m.def("ce_fwd", &ce_fwd, "cross-entropy forward");
   m.def("ce_bwd", &ce_bwd, "cross-entropy backward");
+  m.def("revert_varlen", &revert_varlen,
+        "reverse each batch element's valid time prefix (T,B,F)");
+  m.def("varlen_mask", &varlen_mask, "valid-timestep mask (T,B) uint8");
   m.def("dw3x3_fwd", &dw3x3_fwd, "depthwise 3x3 forward");
   m.def("dw3x3_bwd", &dw3x3_bwd, "depthwise 3x3 backward");
 }
