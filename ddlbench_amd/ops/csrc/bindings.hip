@@ -14,6 +14,10 @@
 template <typename T>
 void launch_fused_sgd(uintptr_t*, uintptr_t*, uintptr_t*, const int64_t*,
                       int, int64_t, float, float, float, int, hipStream_t);
+template <typename T>
+void launch_fused_adam(uintptr_t*, uintptr_t*, uintptr_t*, uintptr_t*,
+                       const int64_t*, int, int64_t, float, float, float,
+                       float, float, float, float, int, hipStream_t);
 
 template <typename T>
 void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int,
@@ -106,6 +110,30 @@ void fused_sgd(torch::Tensor ptr_params, torch::Tensor ptr_grads,
         dptr<uintptr_t>(ptr_moms), dptr<int64_t>(prefix),
         (int)ptr_params.numel(), total, (float)lr, (float)momentum,
         (float)weight_decay, first_step, s);
+}
+
+void fused_adam(torch::Tensor ptr_params, torch::Tensor ptr_grads,
+                torch::Tensor ptr_ms, torch::Tensor ptr_vs,
+                torch::Tensor prefix, int64_t total, double lr,
+                double beta1, double beta2, double eps,
+                double weight_decay, double bc1, double bc2,
+                bool decoupled_wd, bool bf16) {
+  check_gpu_contig(ptr_params, "ptr_params");
+  auto s = cur_stream();
+  if (bf16)
+    launch_fused_adam<__hip_bfloat16>(
+        dptr<uintptr_t>(ptr_params), dptr<uintptr_t>(ptr_grads),
+        dptr<uintptr_t>(ptr_ms), dptr<uintptr_t>(ptr_vs),
+        dptr<int64_t>(prefix), (int)ptr_params.numel(), total, (float)lr,
+        (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+        (float)bc1, (float)bc2, decoupled_wd, s);
+  else
+    launch_fused_adam<float>(
+        dptr<uintptr_t>(ptr_params), dptr<uintptr_t>(ptr_grads),
+        dptr<uintptr_t>(ptr_ms), dptr<uintptr_t>(ptr_vs),
+        dptr<int64_t>(prefix), (int)ptr_params.numel(), total, (float)lr,
+        (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+        (float)bc1, (float)bc2, decoupled_wd, s);
 }
 
 // ---- fused BN + act (+residual) ---------------------------------------
@@ -380,6 +408,7 @@ torch::Tensor varlen_mask(torch::Tensor lengths, int64_t T) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "ddlbench_amd gfx950 HIP kernels";
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD step");
+  m.def("fused_adam", &fused_adam, "fused multi-tensor Adam/AdamW step");
   m.def("bn_act_fwd", &bn_act_fwd, "fused BN+act(+res) forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act(+res) backward");
   m.def("conv_igemm_fwd", &conv_igemm_fwd,

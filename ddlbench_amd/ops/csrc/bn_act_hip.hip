@@ -64,26 +64,55 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, double* __restrict__ su
   }
 }
 
-// ---- stats, NHWC (channels-last): thread-per-channel, coalesced -------
-// rows = N*H*W; element (row, c) at row*C + c. Consecutive lanes read
-// consecutive channels -> every row access is one coalesced segment.
+// ---- stats, NHWC (channels-last) --------------------------------------
+// rows = N*H*W; element (row, c) at row*C + c. Block = CG channels x
+// (256/CG) row-groups: every wave reads consecutive channels of one row
+// (coalesced), the row-groups + a 4x unroll keep >=16 rows in flight per
+// block; per-channel partials combine across row-groups in LDS, one f64
+// atomic per channel per block.
 template <typename T>
 __global__ void bn_stats_nhwc_kernel(const T* __restrict__ x,
                                      double* __restrict__ sums,
-                                     int64_t rows, int64_t C) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                     int64_t rows, int64_t C, int CG) {
+  __shared__ double tmp[2 * 256];
+  const int ci = threadIdx.x % CG;
+  const int rj = threadIdx.x / CG;
+  const int RG = blockDim.x / CG;
+  const int64_t c = (int64_t)blockIdx.x * CG + ci;
+  const bool active = (c < C) && (rj < RG);
   const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
   const int64_t begin = (int64_t)blockIdx.y * per;
   const int64_t end = i64min(begin + per, rows);
   double s = 0.0, ss = 0.0;
-  for (int64_t r = begin; r < end; ++r) {
-    const float v = to_f32(x[r * C + c]);
-    s += v;
-    ss += fma((double)v, (double)v, 0.0);
+  if (active) {
+    int64_t r = begin + rj;
+    const int64_t step = RG;
+    for (; r + 3 * step < end; r += 4 * step) {
+      float v0 = to_f32(x[(r + 0 * step) * C + c]);
+      float v1 = to_f32(x[(r + 1 * step) * C + c]);
+      float v2 = to_f32(x[(r + 2 * step) * C + c]);
+      float v3 = to_f32(x[(r + 3 * step) * C + c]);
+      s += (double)v0 + v1 + v2 + v3;
+      ss += fma((double)v0, v0, fma((double)v1, v1,
+                fma((double)v2, v2, (double)v3 * v3)));
+    }
+    for (; r < end; r += step) {
+      const float v = to_f32(x[r * C + c]);
+      s += v;
+      ss += fma((double)v, (double)v, 0.0);
+    }
   }
-  atomicAdd(&sums[c], s);
-  atomicAdd(&sums[C + c], ss);
+  tmp[threadIdx.x] = s;
+  tmp[256 + threadIdx.x] = ss;
+  __syncthreads();
+  if (rj == 0 && c < C) {
+    for (int j = 1; j < RG; ++j) {
+      s += tmp[j * CG + ci];
+      ss += tmp[256 + j * CG + ci];
+    }
+    atomicAdd(&sums[c], s);
+    atomicAdd(&sums[C + c], ss);
+  }
 }
 
 // ---- finalize: mean/invstd + running-stat update ----------------------
@@ -128,6 +157,121 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
   }
 }
 
+// ---- vectorized apply / dx: 8 elements (16 B) per thread -------------
+// NHWC: 8 consecutive elements are 8 consecutive channels (chunks are
+// 8-aligned in c when C %% 8 == 0); NCHW: 8 consecutive elements share
+// one channel when HW %% 8 == 0. Dispatcher falls back to the scalar
+// kernels otherwise.
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+template <typename T> struct vec8;
+template <> struct vec8<float> {
+  using type = __attribute__((ext_vector_type(8))) float;
+};
+template <> struct vec8<__hip_bfloat16> { using type = short8v; };
+
+DEV float elt_f32(const short8v& v, int j) {
+  __hip_bfloat16 h;
+  unsigned short u = (unsigned short)v[j];
+  __builtin_memcpy(&h, &u, 2);
+  return __bfloat162float(h);
+}
+DEV float elt_f32(const __attribute__((ext_vector_type(8))) float& v,
+                  int j) { return v[j]; }
+
+template <typename T>
+DEV void set_elt(short8v& v, int j, float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  unsigned short u;
+  __builtin_memcpy(&u, &h, 2);
+  v[j] = (short)u;
+}
+template <typename T>
+DEV void set_elt(__attribute__((ext_vector_type(8))) float& v, int j,
+                 float f) { v[j] = f; }
+
+template <typename T, int ACT, bool ADD, bool NHWC>
+__global__ void bn_apply_vec_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ res,
+                                    T* __restrict__ y,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ beta,
+                                    int64_t C, int64_t cdiv,
+                                    int64_t total8) {
+  using V = typename vec8<T>::type;
+  const V* xv = reinterpret_cast<const V*>(x);
+  const V* rv = reinterpret_cast<const V*>(res);
+  V* yv = reinterpret_cast<V*>(y);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total8; i += stride) {
+    const int64_t e0 = i * 8;
+    V vx = xv[i];
+    V vr;
+    if (ADD) vr = rv[i];
+    V vy;
+    if (NHWC) {
+      const int64_t c0 = e0 % C;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int64_t c = c0 + j;
+        float v = (elt_f32(vx, j) - mean[c]) * invstd[c] * gamma[c]
+                  + beta[c];
+        if (ADD) v += elt_f32(vr, j);
+        set_elt<T>(vy, j, act_fwd<ACT>(v));
+      }
+    } else {
+      const int64_t c = (e0 / cdiv) % C;
+      const float mu = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = (elt_f32(vx, j) - mu) * is * g + b;
+        if (ADD) v += elt_f32(vr, j);
+        set_elt<T>(vy, j, act_fwd<ACT>(v));
+      }
+    }
+    yv[i] = vy;
+  }
+}
+
+template <typename T, int ACT, bool ADD, bool NHWC>
+__global__ void bn_bwd_dx_vec_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ y,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     const float* __restrict__ k,
+                                     T* __restrict__ dx,
+                                     T* __restrict__ dres, int64_t C,
+                                     int64_t cdiv, int64_t total8) {
+  using V = typename vec8<T>::type;
+  const V* dyv = reinterpret_cast<const V*>(dy);
+  const V* yv = reinterpret_cast<const V*>(y);
+  const V* xv = reinterpret_cast<const V*>(x);
+  V* dxv = reinterpret_cast<V*>(dx);
+  V* drv = reinterpret_cast<V*>(dres);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total8; i += stride) {
+    const int64_t e0 = i * 8;
+    V vdy = dyv[i], vy = yv[i], vx = xv[i];
+    V vdx, vdr;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int64_t c = NHWC ? (e0 % C + j) : ((e0 / cdiv) % C);
+      const float g = elt_f32(vdy, j) * act_mask<ACT>(elt_f32(vy, j));
+      const float xhat = (elt_f32(vx, j) - mean[c]) * invstd[c];
+      set_elt<T>(vdx, j, k[c] * (g - k[C + c] - xhat * k[2 * C + c]));
+      if (ADD) set_elt<T>(vdr, j, g);
+    }
+    dxv[i] = vdx;
+    if (ADD) drv[i] = vdr;
+  }
+}
+
 // ---- backward reduce: per-channel Σdy', Σdy'*xhat ---------------------
 template <typename T, int ACT>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
@@ -163,7 +307,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
   }
 }
 
-// ---- backward reduce, NHWC: thread-per-channel, coalesced -------------
+// ---- backward reduce, NHWC (same block geometry as the stats) ---------
 template <typename T, int ACT>
 __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
                                           const T* __restrict__ y,
@@ -171,23 +315,38 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
                                           const float* __restrict__ mean,
                                           const float* __restrict__ invstd,
                                           double* __restrict__ sums,
-                                          int64_t rows, int64_t C) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float mu = mean[c], is = invstd[c];
+                                          int64_t rows, int64_t C, int CG) {
+  __shared__ double tmp[2 * 256];
+  const int ci = threadIdx.x % CG;
+  const int rj = threadIdx.x / CG;
+  const int RG = blockDim.x / CG;
+  const int64_t c = (int64_t)blockIdx.x * CG + ci;
+  const bool active = (c < C) && (rj < RG);
   const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
   const int64_t begin = (int64_t)blockIdx.y * per;
   const int64_t end = i64min(begin + per, rows);
   double sdy = 0.0, sdyx = 0.0;
-  for (int64_t r = begin; r < end; ++r) {
-    const int64_t idx = r * C + c;
-    const float g = to_f32(dy[idx]) * act_mask<ACT>(to_f32(y[idx]));
-    const float xhat = (to_f32(x[idx]) - mu) * is;
-    sdy += g;
-    sdyx += fma((double)g, (double)xhat, 0.0);
+  if (active) {
+    const float mu = mean[c], is = invstd[c];
+    for (int64_t r = begin + rj; r < end; r += RG) {
+      const int64_t idx = r * C + c;
+      const float g = to_f32(dy[idx]) * act_mask<ACT>(to_f32(y[idx]));
+      const float xhat = (to_f32(x[idx]) - mu) * is;
+      sdy += g;
+      sdyx += fma((double)g, (double)xhat, 0.0);
+    }
   }
-  atomicAdd(&sums[c], sdy);
-  atomicAdd(&sums[C + c], sdyx);
+  tmp[threadIdx.x] = sdy;
+  tmp[256 + threadIdx.x] = sdyx;
+  __syncthreads();
+  if (rj == 0 && c < C) {
+    for (int j = 1; j < RG; ++j) {
+      sdy += tmp[j * CG + ci];
+      sdyx += tmp[256 + j * CG + ci];
+    }
+    atomicAdd(&sums[c], sdy);
+    atomicAdd(&sums[C + c], sdyx);
+  }
 }
 
 __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
@@ -241,10 +400,11 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    const int64_t cblocks = (C + block - 1) / block;
-    int64_t S = i64min(i64max(rows / 1024, 1), i64max(2048 / cblocks, 1));
+    const int CG = C >= 64 ? 64 : (int)C;
+    const int64_t cblocks = (C + CG - 1) / CG;
+    int64_t S = i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
     hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
-                       dim3(block), 0, stream, x, sums, rows, C);
+                       dim3(block), 0, stream, x, sums, rows, C, CG);
   } else {
     int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
     S = i64max(S, 1);
@@ -270,8 +430,31 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
                      const float* beta, int64_t C, int64_t HW, int64_t total,
                      int act, int nhwc, hipStream_t stream) {
   const int block = 256;
-  const int grid = elementwise_grid(total, block);
   const int64_t cdiv = nhwc ? 1 : HW;
+  const bool vec8ok = (total % 8 == 0) &&
+                      (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
+  if (vec8ok) {
+    const int64_t total8 = total / 8;
+    const int grid8 = elementwise_grid(total8, block);
+    const bool add = res != nullptr;
+#define VCASE(ACT, ADD, NHWC)                                               \
+    hipLaunchKernelGGL((bn_apply_vec_kernel<T, ACT, ADD, NHWC>),            \
+                       dim3(grid8), dim3(block), 0, stream, x, res, y,      \
+                       mean, invstd, gamma, beta, C, cdiv, total8)
+#define VSEL(ACT)                                                           \
+    do { if (nhwc) { if (add) VCASE(ACT, true, true);                       \
+                     else VCASE(ACT, false, true); }                        \
+         else { if (add) VCASE(ACT, true, false);                           \
+                else VCASE(ACT, false, false); } } while (0)
+    if (act == 0) VSEL(0);
+    else if (act == 1) VSEL(1);
+    else VSEL(2);
+#undef VSEL
+#undef VCASE
+    HIP_CHECK_LAST();
+    return;
+  }
+  const int grid = elementwise_grid(total, block);
 #define CASE(ACT, ADD)                                                       \
   hipLaunchKernelGGL((bn_apply_kernel<T, ACT, ADD>), dim3(grid), dim3(block), \
                      0, stream, x, res, y, mean, invstd, gamma, beta, C,    \
@@ -295,12 +478,13 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    const int64_t cblocks = (C + block - 1) / block;
-    int64_t S = i64min(i64max(rows / 1024, 1), i64max(2048 / cblocks, 1));
+    const int CG = C >= 64 ? 64 : (int)C;
+    const int64_t cblocks = (C + CG - 1) / CG;
+    int64_t S = i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
 #define CASE(ACT)                                                           \
     hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT>),                 \
                        dim3(cblocks, S), dim3(block), 0, stream, dy, y, x,  \
-                       mean, invstd, sums, rows, C)
+                       mean, invstd, sums, rows, C, CG)
     if (act == 0) CASE(0);
     else if (act == 1) CASE(1);
     else CASE(2);
@@ -337,8 +521,31 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
                       int64_t C, int64_t HW, int64_t total, int act,
                       int nhwc, hipStream_t stream) {
   const int block = 256;
-  const int grid = elementwise_grid(total, block);
   const int64_t cdiv = nhwc ? 1 : HW;
+  const bool vec8ok = (total % 8 == 0) &&
+                      (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
+  if (vec8ok) {
+    const int64_t total8 = total / 8;
+    const int grid8 = elementwise_grid(total8, block);
+    const bool add = dres != nullptr;
+#define VCASE(ACT, ADD, NHWC)                                               \
+    hipLaunchKernelGGL((bn_bwd_dx_vec_kernel<T, ACT, ADD, NHWC>),           \
+                       dim3(grid8), dim3(block), 0, stream, dy, y, x,       \
+                       mean, invstd, k, dx, dres, C, cdiv, total8)
+#define VSEL(ACT)                                                           \
+    do { if (nhwc) { if (add) VCASE(ACT, true, true);                       \
+                     else VCASE(ACT, false, true); }                        \
+         else { if (add) VCASE(ACT, true, false);                           \
+                else VCASE(ACT, false, false); } } while (0)
+    if (act == 0) VSEL(0);
+    else if (act == 1) VSEL(1);
+    else VSEL(2);
+#undef VSEL
+#undef VCASE
+    HIP_CHECK_LAST();
+    return;
+  }
+  const int grid = elementwise_grid(total, block);
 #define CASE(ACT, ADD)                                                    \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<T, ACT, ADD>), dim3(grid),         \
                      dim3(block), 0, stream, dy, y, x, mean, invstd, k,   \

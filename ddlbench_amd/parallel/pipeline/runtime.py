@@ -83,6 +83,25 @@ class StagePlan:
         return max(after // self.replicas[stage], 0)
 
 
+class RuntimeStats:
+    """fwd/bwd counters: compute time, send/recv bytes (the reference's
+    runtime_utilities.RuntimeStats)."""
+
+    def __init__(self) -> None:
+        self.reset()
+
+    def reset(self) -> None:
+        self.fwd_count = 0
+        self.bwd_count = 0
+        self.send_bytes = 0
+        self.recv_bytes = 0
+
+    def as_dict(self) -> dict:
+        return {"fwd_count": self.fwd_count, "bwd_count": self.bwd_count,
+                "send_bytes": self.send_bytes,
+                "recv_bytes": self.recv_bytes}
+
+
 class StageRuntime:
     def __init__(self, plan: StagePlan, rank: int,
                  module: torch.nn.Module,
@@ -108,6 +127,7 @@ class StageRuntime:
         self.is_last = self.stage == plan.num_stages - 1
         # in-flight state: (x, out_or_loss, send_work, mb_index, extras)
         self.inflight: deque = deque()
+        self.stats = RuntimeStats()
 
     # ---- routing helpers ----------------------------------------------
     def _prev_rank(self, mb: int) -> int:
@@ -134,6 +154,7 @@ class StageRuntime:
             ch = self.transport.channel(self._prev_rank(mb), self.rank,
                                         "fwd")
             ch.irecv(buf).wait()
+            self.stats.recv_bytes += buf.numel() * buf.element_size()
             x = buf
         if training and not self.is_first:
             x.requires_grad_(True)
@@ -152,8 +173,10 @@ class StageRuntime:
             ch = self.transport.channel(self.rank, self._next_rank(mb),
                                         "fwd")
             send_work = ch.isend(out.detach())
+            self.stats.send_bytes += out.numel() * out.element_size()
         self.inflight.append((x, loss if self.is_last else out, send_work,
                               mb, extras))
+        self.stats.fwd_count += 1
         return loss, extras
 
     def run_backward(self):
@@ -176,6 +199,8 @@ class StageRuntime:
             # is safe (upstream is blocked on it anyway — the pipeline's
             # critical path is unaffected)
             ch.isend(x.grad).wait()
+            self.stats.send_bytes += x.grad.numel() * x.grad.element_size()
+        self.stats.bwd_count += 1
         return mb
 
     def pop_eval(self):
