@@ -114,6 +114,110 @@ __global__ void bn_stats_nhwc_kernel(const T* __restrict__ x,
   }
 }
 
+// ---- stats, NHWC vectorized: 8 channels (16 B) per lane --------------
+// Each thread covers channels c0..c0+7 of CG8*8-channel groups; f32
+// per-thread partials over a short row slice (<= a few hundred values)
+// combine in LDS across row-groups, then one f64 atomic per channel.
+template <typename T>
+__global__ void bn_stats_nhwc_vec_kernel(const T* __restrict__ x,
+                                         double* __restrict__ sums,
+                                         int64_t rows, int64_t C, int CG8) {
+  typedef __attribute__((ext_vector_type(8))) short short8x;
+  __shared__ float tmp[2][256][8];
+  const int ci = threadIdx.x % CG8;
+  const int rj = threadIdx.x / CG8;
+  const int RG = blockDim.x / CG8;
+  const int64_t c0 = ((int64_t)blockIdx.x * CG8 + ci) * 8;
+  const bool active = (c0 < C) && (rj < RG);
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (active) {
+    const short8x* xv = reinterpret_cast<const short8x*>(x);
+    for (int64_t r = begin + rj; r < end; r += RG) {
+      short8x v = xv[(r * C + c0) / 8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __hip_bfloat16 h;
+        unsigned short u = (unsigned short)v[j];
+        __builtin_memcpy(&h, &u, 2);
+        const float f = __bfloat162float(h);
+        s[j] += f;
+        ss[j] = fmaf(f, f, ss[j]);
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    tmp[0][threadIdx.x][j] = s[j];
+    tmp[1][threadIdx.x][j] = ss[j];
+  }
+  __syncthreads();
+  if (rj == 0 && c0 < C) {
+    for (int g = 1; g < RG; ++g)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s[j] += tmp[0][g * CG8 + ci][j];
+        ss[j] += tmp[1][g * CG8 + ci][j];
+      }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sums[c0 + j], (double)s[j]);
+      atomicAdd(&sums[C + c0 + j], (double)ss[j]);
+    }
+  }
+}
+
+// fp32 NHWC vec stats (4 channels / 16 B)
+__global__ void bn_stats_nhwc_vec_f32_kernel(const float* __restrict__ x,
+                                             double* __restrict__ sums,
+                                             int64_t rows, int64_t C,
+                                             int CG4) {
+  typedef __attribute__((ext_vector_type(4))) float float4x;
+  __shared__ float tmp[2][256][4];
+  const int ci = threadIdx.x % CG4;
+  const int rj = threadIdx.x / CG4;
+  const int RG = blockDim.x / CG4;
+  const int64_t c0 = ((int64_t)blockIdx.x * CG4 + ci) * 4;
+  const bool active = (c0 < C) && (rj < RG);
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  float s[4] = {0, 0, 0, 0}, ss[4] = {0, 0, 0, 0};
+  if (active) {
+    const float4x* xv = reinterpret_cast<const float4x*>(x);
+    for (int64_t r = begin + rj; r < end; r += RG) {
+      float4x v = xv[(r * C + c0) / 4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        s[j] += v[j];
+        ss[j] = fmaf(v[j], v[j], ss[j]);
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    tmp[0][threadIdx.x][j] = s[j];
+    tmp[1][threadIdx.x][j] = ss[j];
+  }
+  __syncthreads();
+  if (rj == 0 && c0 < C) {
+    for (int g = 1; g < RG; ++g)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        s[j] += tmp[0][g * CG4 + ci][j];
+        ss[j] += tmp[1][g * CG4 + ci][j];
+      }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      atomicAdd(&sums[c0 + j], (double)s[j]);
+      atomicAdd(&sums[C + c0 + j], (double)ss[j]);
+    }
+  }
+}
+
 // ---- finalize: mean/invstd + running-stat update ----------------------
 __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ mean,
@@ -348,6 +452,77 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
   }
 }
 
+template <typename T, int ACT>
+__global__ void bn_bwd_reduce_nhwc_vec_kernel(
+    const T* __restrict__ dy, const T* __restrict__ y,
+    const T* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, double* __restrict__ sums,
+    int64_t rows, int64_t C, int CG8) {
+  typedef __attribute__((ext_vector_type(8))) short short8x;
+  __shared__ float tmp[2][256][8];
+  const int ci = threadIdx.x % CG8;
+  const int rj = threadIdx.x / CG8;
+  const int RG = blockDim.x / CG8;
+  const int64_t c0 = ((int64_t)blockIdx.x * CG8 + ci) * 8;
+  const bool active = (c0 < C) && (rj < RG);
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  float sdy[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float sdyx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float mu[8], is[8];
+  if (active) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[c0 + j];
+      is[j] = invstd[c0 + j];
+    }
+    const short8x* dyv = reinterpret_cast<const short8x*>(dy);
+    const short8x* yv = reinterpret_cast<const short8x*>(y);
+    const short8x* xv = reinterpret_cast<const short8x*>(x);
+    for (int64_t r = begin + rj; r < end; r += RG) {
+      const int64_t i8 = (r * C + c0) / 8;
+      short8x vdy = dyv[i8], vy = yv[i8], vx = xv[i8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __hip_bfloat16 h;
+        unsigned short u;
+        u = (unsigned short)vdy[j];
+        __builtin_memcpy(&h, &u, 2);
+        const float fdy = __bfloat162float(h);
+        u = (unsigned short)vy[j];
+        __builtin_memcpy(&h, &u, 2);
+        const float fy = __bfloat162float(h);
+        u = (unsigned short)vx[j];
+        __builtin_memcpy(&h, &u, 2);
+        const float fx = __bfloat162float(h);
+        const float g = fdy * act_mask<ACT>(fy);
+        sdy[j] += g;
+        sdyx[j] = fmaf(g, (fx - mu[j]) * is[j], sdyx[j]);
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    tmp[0][threadIdx.x][j] = sdy[j];
+    tmp[1][threadIdx.x][j] = sdyx[j];
+  }
+  __syncthreads();
+  if (rj == 0 && c0 < C) {
+    for (int g = 1; g < RG; ++g)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sdy[j] += tmp[0][g * CG8 + ci][j];
+        sdyx[j] += tmp[1][g * CG8 + ci][j];
+      }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sums[c0 + j], (double)sdy[j]);
+      atomicAdd(&sums[C + c0 + j], (double)sdyx[j]);
+    }
+  }
+}
+
 __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ invstd,
@@ -399,11 +574,29 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    const int CG = C >= 64 ? 64 : (int)C;
-    const int64_t cblocks = (C + CG - 1) / CG;
-    int64_t S = i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
-    hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
-                       dim3(block), 0, stream, x, sums, rows, C, CG);
+    if (sizeof(T) == 2 && C % 8 == 0) {
+      const int CG8 = (int)i64min(C / 8, 64);
+      const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
+      int64_t S = i64min(i64max(rows / 512, 1),
+                         i64max(2048 / cblocks, 1));
+      hipLaunchKernelGGL((bn_stats_nhwc_vec_kernel<T>), dim3(cblocks, S),
+                         dim3(block), 0, stream, x, sums, rows, C, CG8);
+    } else if (sizeof(T) == 4 && C % 4 == 0) {
+      const int CG4 = (int)i64min(C / 4, 64);
+      const int64_t cblocks = (C / 4 + CG4 - 1) / CG4;
+      int64_t S = i64min(i64max(rows / 512, 1),
+                         i64max(2048 / cblocks, 1));
+      hipLaunchKernelGGL(bn_stats_nhwc_vec_f32_kernel, dim3(cblocks, S),
+                         dim3(block), 0, stream, (const float*)x, sums,
+                         rows, C, CG4);
+    } else {
+      const int CG = C >= 64 ? 64 : (int)C;
+      const int64_t cblocks = (C + CG - 1) / CG;
+      int64_t S = i64min(i64max(rows / 512, 1),
+                         i64max(2048 / cblocks, 1));
+      hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
+                         dim3(block), 0, stream, x, sums, rows, C, CG);
+    }
   } else {
     int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
     S = i64max(S, 1);
@@ -477,17 +670,33 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    const int CG = C >= 64 ? 64 : (int)C;
-    const int64_t cblocks = (C + CG - 1) / CG;
-    int64_t S = i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
+    if (sizeof(T) == 2 && C % 8 == 0) {
+      const int CG8 = (int)i64min(C / 8, 64);
+      const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
+      int64_t S = i64min(i64max(rows / 512, 1),
+                         i64max(2048 / cblocks, 1));
 #define CASE(ACT)                                                           \
-    hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT>),                 \
-                       dim3(cblocks, S), dim3(block), 0, stream, dy, y, x,  \
-                       mean, invstd, sums, rows, C, CG)
-    if (act == 0) CASE(0);
-    else if (act == 1) CASE(1);
-    else CASE(2);
+      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_vec_kernel<T, ACT>),           \
+                         dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
+                         x, mean, invstd, sums, rows, C, CG8)
+      if (act == 0) CASE(0);
+      else if (act == 1) CASE(1);
+      else CASE(2);
 #undef CASE
+    } else {
+      const int CG = C >= 64 ? 64 : (int)C;
+      const int64_t cblocks = (C + CG - 1) / CG;
+      int64_t S = i64min(i64max(rows / 512, 1),
+                         i64max(2048 / cblocks, 1));
+#define CASE(ACT)                                                           \
+      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT>),               \
+                         dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
+                         x, mean, invstd, sums, rows, C, CG)
+      if (act == 0) CASE(0);
+      else if (act == 1) CASE(1);
+      else CASE(2);
+#undef CASE
+    }
   } else {
     int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
     S = i64max(S, 1);
