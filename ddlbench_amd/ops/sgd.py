@@ -104,9 +104,16 @@ class FusedSGD(Optimizer):
                 # channels_last) as long as grad strides match param's
                 assert p.grad.stride() == p.stride(), \
                     "FusedSGD needs grads with the param's layout"
-            ptr_grads = torch.tensor([p.grad.data_ptr() for p in ps],
-                                     dtype=torch.int64).to(dev,
-                                                           non_blocking=True)
+            # grads in persistent buckets (DP) keep their pointers —
+            # re-upload the pointer array only when one moved
+            gptrs = [p.grad.data_ptr() for p in ps]
+            if cached.get("grad_ptrs_host") == gptrs:
+                ptr_grads = cached["ptr_grads"]
+            else:
+                ptr_grads = torch.tensor(gptrs, dtype=torch.int64).to(
+                    dev, non_blocking=True)
+                cached["grad_ptrs_host"] = gptrs
+                cached["ptr_grads"] = ptr_grads
             ext.fused_sgd(cached["ptr_params"], ptr_grads,
                           cached["ptr_moms"], cached["prefix"],
                           cached["total"], lr, mu, wd, first_step,
