@@ -376,6 +376,130 @@ __global__ void bn_bwd_dx_vec_kernel(const T* __restrict__ dy,
   }
 }
 
+// ---- NHWC apply / dx, fixed-channel geometry ---------------------------
+// Same (CG8 x RG) block shape as the reductions: each thread owns 8
+// channels and walks rows, so per-channel params load ONCE per thread
+// instead of once per element (the grid-stride variant is issue-bound
+// on the 32 extra scalar loads per 48 B of traffic).
+template <typename T, int ACT, bool ADD>
+__global__ void bn_apply_nhwc_kernel2(const T* __restrict__ x,
+                                      const T* __restrict__ res,
+                                      T* __restrict__ y,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ beta,
+                                      int64_t rows, int64_t C, int CG8) {
+  typedef __attribute__((ext_vector_type(8))) short short8x;
+  const int ci = threadIdx.x % CG8;
+  const int rj = threadIdx.x / CG8;
+  const int RG = blockDim.x / CG8;
+  const int64_t c0 = ((int64_t)blockIdx.x * CG8 + ci) * 8;
+  if (c0 >= C || rj >= RG) return;
+  float sc[8], sh[8], rs[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = invstd[c0 + j] * gamma[c0 + j];
+    sh[j] = beta[c0 + j] - mean[c0 + j] * sc[j];
+  }
+  (void)rs;
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  const short8x* xv = reinterpret_cast<const short8x*>(x);
+  const short8x* rv = reinterpret_cast<const short8x*>(res);
+  short8x* yv = reinterpret_cast<short8x*>(y);
+  for (int64_t r = begin + rj; r < end; r += RG) {
+    const int64_t i8 = (r * C + c0) / 8;
+    short8x vx = xv[i8];
+    short8x vr;
+    if (ADD) vr = rv[i8];
+    short8x vy;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_bfloat16 h;
+      unsigned short u = (unsigned short)vx[j];
+      __builtin_memcpy(&h, &u, 2);
+      float v = fmaf(__bfloat162float(h), sc[j], sh[j]);
+      if (ADD) {
+        u = (unsigned short)vr[j];
+        __builtin_memcpy(&h, &u, 2);
+        v += __bfloat162float(h);
+      }
+      h = __float2bfloat16(act_fwd<ACT>(v));
+      __builtin_memcpy(&u, &h, 2);
+      vy[j] = (short)u;
+    }
+    yv[i8] = vy;
+  }
+}
+
+template <typename T, int ACT, bool ADD>
+__global__ void bn_bwd_dx_nhwc_kernel2(const T* __restrict__ dy,
+                                       const T* __restrict__ y,
+                                       const T* __restrict__ x,
+                                       const float* __restrict__ mean,
+                                       const float* __restrict__ invstd,
+                                       const float* __restrict__ k,
+                                       T* __restrict__ dx,
+                                       T* __restrict__ dres, int64_t rows,
+                                       int64_t C, int CG8) {
+  typedef __attribute__((ext_vector_type(8))) short short8x;
+  const int ci = threadIdx.x % CG8;
+  const int rj = threadIdx.x / CG8;
+  const int RG = blockDim.x / CG8;
+  const int64_t c0 = ((int64_t)blockIdx.x * CG8 + ci) * 8;
+  if (c0 >= C || rj >= RG) return;
+  float k1[8], k2[8], k3[8], mu[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    k1[j] = k[c0 + j];
+    k2[j] = k[C + c0 + j];
+    k3[j] = k[2 * C + c0 + j];
+    mu[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
+  const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
+  const int64_t begin = (int64_t)blockIdx.y * per;
+  const int64_t end = i64min(begin + per, rows);
+  const short8x* dyv = reinterpret_cast<const short8x*>(dy);
+  const short8x* yv = reinterpret_cast<const short8x*>(y);
+  const short8x* xv = reinterpret_cast<const short8x*>(x);
+  short8x* dxv = reinterpret_cast<short8x*>(dx);
+  short8x* drv = reinterpret_cast<short8x*>(dres);
+  for (int64_t r = begin + rj; r < end; r += RG) {
+    const int64_t i8 = (r * C + c0) / 8;
+    short8x vdy = dyv[i8], vy = yv[i8], vx = xv[i8];
+    short8x vdx, vdr;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_bfloat16 h;
+      unsigned short u;
+      u = (unsigned short)vdy[j];
+      __builtin_memcpy(&h, &u, 2);
+      const float fdy = __bfloat162float(h);
+      u = (unsigned short)vy[j];
+      __builtin_memcpy(&h, &u, 2);
+      const float fy = __bfloat162float(h);
+      u = (unsigned short)vx[j];
+      __builtin_memcpy(&h, &u, 2);
+      const float fx = __bfloat162float(h);
+      const float g = fdy * act_mask<ACT>(fy);
+      const float xhat = (fx - mu[j]) * is[j];
+      h = __float2bfloat16(k1[j] * (g - k2[j] - xhat * k3[j]));
+      __builtin_memcpy(&u, &h, 2);
+      vdx[j] = (short)u;
+      if (ADD) {
+        h = __float2bfloat16(g);
+        __builtin_memcpy(&u, &h, 2);
+        vdr[j] = (short)u;
+      }
+    }
+    dxv[i8] = vdx;
+    if (ADD) drv[i8] = vdr;
+  }
+}
+
 // ---- backward reduce: per-channel Σdy', Σdy'*xhat ---------------------
 template <typename T, int ACT>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
@@ -626,6 +750,23 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0) {
+    const int64_t rows = total / C;
+    const int CG8 = (int)i64min(C / 8, 64);
+    const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
+    int64_t S = i64min(i64max(rows / 256, 1), i64max(2048 / cblocks, 1));
+    const bool add = res != nullptr;
+#define KCASE(ACT, ADD)                                                     \
+    hipLaunchKernelGGL((bn_apply_nhwc_kernel2<T, ACT, ADD>),                \
+                       dim3(cblocks, S), dim3(block), 0, stream, x, res,    \
+                       y, mean, invstd, gamma, beta, rows, C, CG8)
+    if (act == 0) { if (add) KCASE(0, true); else KCASE(0, false); }
+    else if (act == 1) { if (add) KCASE(1, true); else KCASE(1, false); }
+    else { if (add) KCASE(2, true); else KCASE(2, false); }
+#undef KCASE
+    HIP_CHECK_LAST();
+    return;
+  }
   if (vec8ok) {
     const int64_t total8 = total / 8;
     const int grid8 = elementwise_grid(total8, block);
@@ -733,6 +874,23 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0) {
+    const int64_t rows = total / C;
+    const int CG8 = (int)i64min(C / 8, 64);
+    const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
+    int64_t S = i64min(i64max(rows / 256, 1), i64max(2048 / cblocks, 1));
+    const bool add = dres != nullptr;
+#define KCASE(ACT, ADD)                                                     \
+    hipLaunchKernelGGL((bn_bwd_dx_nhwc_kernel2<T, ACT, ADD>),               \
+                       dim3(cblocks, S), dim3(block), 0, stream, dy, y, x,  \
+                       mean, invstd, k, dx, dres, rows, C, CG8)
+    if (act == 0) { if (add) KCASE(0, true); else KCASE(0, false); }
+    else if (act == 1) { if (add) KCASE(1, true); else KCASE(1, false); }
+    else { if (add) KCASE(2, true); else KCASE(2, false); }
+#undef KCASE
+    HIP_CHECK_LAST();
+    return;
+  }
   if (vec8ok) {
     const int64_t total8 = total / 8;
     const int grid8 = elementwise_grid(total8, block);

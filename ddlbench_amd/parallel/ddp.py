@@ -135,7 +135,9 @@ class BucketedDataParallel(nn.Module):
 
     def zero_grad_buckets(self) -> None:
         if self.world_size <= 1:
-            self.module.zero_grad(set_to_none=False)
+            # set_to_none avoids one fill launch per param AND the
+            # read-modify-write grad accumulation pass
+            self.module.zero_grad(set_to_none=True)
             return
         for b in self._buckets:
             b.flat.zero_()

@@ -142,6 +142,25 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
         n_mb = len(ds) // B
         lcm = plan.lcm_replicas()
         n_mb = max((n_mb // lcm) * lcm, 0)
+        if device.type == "cuda":
+            # device-resident synthetic stream: stage 0 (inputs) and the
+            # last stage (targets) derive the SAME minibatch from the
+            # same (seed, epoch, mb) — no CPU dataset, no H2D copies
+            c, h, w = cfg.shape
+
+            def input_provider(mb):
+                g = torch.Generator(device=device).manual_seed(
+                    cfg.seed * 977 + epoch * 65537 + mb)
+                return torch.randn((B, c, h, w), generator=g,
+                                   device=device, dtype=dtype)
+
+            def target_provider(mb):
+                g = torch.Generator(device=device).manual_seed(
+                    cfg.seed * 977 + epoch * 65537 + mb + 1)
+                return torch.randint(cfg.num_classes, (B,), generator=g,
+                                     device=device)
+
+            return n_mb, input_provider, target_provider
         g = torch.Generator().manual_seed(cfg.seed * 977 + epoch)
         perm = torch.randperm(len(ds), generator=g)
 
