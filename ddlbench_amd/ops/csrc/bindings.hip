@@ -22,6 +22,7 @@ void launch_fused_adam(uintptr_t*, uintptr_t*, uintptr_t*, uintptr_t*,
 template <typename T>
 void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int,
                      hipStream_t);
+void set_bn_variant(int v);
 void launch_bn_finalize(const double*, float*, float*, float*, float*,
                         int64_t, double, float, float, hipStream_t);
 template <typename T>
@@ -410,6 +411,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD step");
   m.def("fused_adam", &fused_adam, "fused multi-tensor Adam/AdamW step");
   m.def("bn_act_fwd", &bn_act_fwd, "fused BN+act(+res) forward");
+  m.def("set_bn_variant", &set_bn_variant,
+        "BN kernel variant for A/B probes (0 auto, 1 scalar-group, 2 vec)");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act(+res) backward");
   m.def("conv_igemm_fwd", &conv_igemm_fwd,
         "NHWC bf16 MFMA implicit-GEMM conv forward");

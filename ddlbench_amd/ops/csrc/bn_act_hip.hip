@@ -36,6 +36,12 @@ template <int ACT> DEV float act_mask(float y) {
 
 }  // namespace
 
+// BN kernel-variant override for A/B probes (set via bindings):
+// 0 = auto, 1 = scalar-group reductions + grid-stride vector apply,
+// 2 = vector reductions + fixed-channel apply
+int g_bn_variant = 0;
+void set_bn_variant(int v) { g_bn_variant = v; }
+
 // ---- stats: per-channel sum / sumsq over N*H*W ------------------------
 // grid = (C, S): block (c, s) covers slice s of channel c's N*HW space.
 template <typename T>
@@ -699,7 +705,7 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    if (sizeof(T) == 2 && C % 8 == 0) {
+    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
       int64_t S = i64min(i64max(rows / 512, 1),
@@ -750,7 +756,7 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
-  if (nhwc && sizeof(T) == 2 && C % 8 == 0) {
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
     const int64_t rows = total / C;
     const int CG8 = (int)i64min(C / 8, 64);
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
@@ -812,7 +818,7 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    if (sizeof(T) == 2 && C % 8 == 0) {
+    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
       int64_t S = i64min(i64max(rows / 512, 1),
@@ -874,7 +880,7 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
-  if (nhwc && sizeof(T) == 2 && C % 8 == 0) {
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
     const int64_t rows = total / C;
     const int CG8 = (int)i64min(C / 8, 64);
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
