@@ -705,7 +705,7 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
+    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
       int64_t S = i64min(i64max(rows / 512, 1),
@@ -818,7 +818,7 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
+    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
       int64_t S = i64min(i64max(rows / 512, 1),
