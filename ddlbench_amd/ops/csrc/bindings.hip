@@ -60,14 +60,14 @@ void launch_varlen_mask(const int64_t*, uint8_t*, int64_t, int64_t,
 
 template <typename T>
 void launch_dw3x3_fwd(const T*, const float*, T*, int64_t, int64_t, int64_t,
-                      int64_t, int64_t, int64_t, int, hipStream_t);
+                      int64_t, int64_t, int64_t, int, int, hipStream_t);
 template <typename T>
 void launch_dw3x3_bwd_dx(const T*, const float*, T*, int64_t, int64_t,
-                         int64_t, int64_t, int64_t, int64_t, int,
+                         int64_t, int64_t, int64_t, int64_t, int, int,
                          hipStream_t);
 template <typename T>
 void launch_dw3x3_bwd_dw(const T*, const T*, double*, int64_t, int64_t,
-                         int64_t, int64_t, int64_t, int64_t, int,
+                         int64_t, int64_t, int64_t, int64_t, int, int,
                          hipStream_t);
 
 namespace {
@@ -328,26 +328,31 @@ torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
 }
 
 // ---- depthwise 3x3 ----------------------------------------------------
-torch::Tensor dw3x3_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
-  check_gpu_contig(x, "x");
+torch::Tensor dw3x3_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                        bool nhwc) {
+  TORCH_CHECK(x.is_cuda(), "x must be on the HIP device");
   check_gpu_contig(w, "w");
   const int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int64_t OH = (H + 2 - 3) / stride + 1, OW = (W + 2 - 3) / stride + 1;
   auto s = cur_stream();
-  torch::Tensor y = torch::empty({N, C, OH, OW}, x.options());
+  torch::Tensor y = nhwc
+      ? torch::empty({N, C, OH, OW}, x.options().memory_format(
+            at::MemoryFormat::ChannelsLast))
+      : torch::empty({N, C, OH, OW}, x.options());
   if (is_bf16(x))
     launch_dw3x3_fwd<__hip_bfloat16>(dptr<__hip_bfloat16>(x), dptr<float>(w),
                                      dptr<__hip_bfloat16>(y), N, C, H, W, OH,
-                                     OW, (int)stride, s);
+                                     OW, (int)stride, nhwc, s);
   else
     launch_dw3x3_fwd<float>(dptr<float>(x), dptr<float>(w), dptr<float>(y),
-                            N, C, H, W, OH, OW, (int)stride, s);
+                            N, C, H, W, OH, OW, (int)stride, nhwc, s);
   return y;
 }
 
 std::vector<torch::Tensor> dw3x3_bwd(torch::Tensor x, torch::Tensor w,
                                      torch::Tensor dy, int64_t stride,
-                                     bool need_dx, bool need_dw) {
+                                     bool need_dx, bool need_dw,
+                                     bool nhwc) {
   const int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int64_t OH = dy.size(2), OW = dy.size(3);
   auto s = cur_stream();
@@ -357,11 +362,11 @@ std::vector<torch::Tensor> dw3x3_bwd(torch::Tensor x, torch::Tensor w,
     if (is_bf16(x))
       launch_dw3x3_bwd_dx<__hip_bfloat16>(
           dptr<__hip_bfloat16>(dy), dptr<float>(w), dptr<__hip_bfloat16>(dx),
-          N, C, H, W, OH, OW, (int)stride, s);
+          N, C, H, W, OH, OW, (int)stride, nhwc, s);
     else
       launch_dw3x3_bwd_dx<float>(dptr<float>(dy), dptr<float>(w),
                                  dptr<float>(dx), N, C, H, W, OH, OW,
-                                 (int)stride, s);
+                                 (int)stride, nhwc, s);
   }
   if (need_dw) {
     torch::Tensor acc = torch::zeros({C * 9}, x.options().dtype(at::kDouble));
@@ -369,11 +374,11 @@ std::vector<torch::Tensor> dw3x3_bwd(torch::Tensor x, torch::Tensor w,
       launch_dw3x3_bwd_dw<__hip_bfloat16>(dptr<__hip_bfloat16>(x),
                                           dptr<__hip_bfloat16>(dy),
                                           dptr<double>(acc), N, C, H, W, OH,
-                                          OW, (int)stride, s);
+                                          OW, (int)stride, nhwc, s);
     else
       launch_dw3x3_bwd_dw<float>(dptr<float>(x), dptr<float>(dy),
                                  dptr<double>(acc), N, C, H, W, OH, OW,
-                                 (int)stride, s);
+                                 (int)stride, nhwc, s);
     dwt = acc.to(at::kFloat).view({C, 1, 3, 3});
   }
   return {dx, dwt};

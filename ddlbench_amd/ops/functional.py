@@ -116,12 +116,15 @@ class _FusedDWConv3x3(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride):
         ext = _ops.require_extension()
-        x = x.contiguous()
+        nhwc = _is_nhwc(x) and x.dtype == torch.bfloat16 \
+            and x.size(1) % 8 == 0
+        x = _layout_contiguous(x, nhwc)
         w32 = weight.detach().float().contiguous()
-        y = ext.dw3x3_fwd(x, w32, stride)
+        y = ext.dw3x3_fwd(x, w32, stride, nhwc)
         ctx.save_for_backward(x, w32)
         ctx.stride = stride
         ctx.wdtype = weight.dtype
+        ctx.nhwc = nhwc
         return y
 
     @staticmethod
@@ -129,8 +132,9 @@ class _FusedDWConv3x3(torch.autograd.Function):
         ext = _ops.require_extension()
         x, w32 = ctx.saved_tensors
         need_dx, need_dw = ctx.needs_input_grad[0], ctx.needs_input_grad[1]
-        dx, dw = ext.dw3x3_bwd(x, w32, dy.contiguous(), ctx.stride,
-                               need_dx, need_dw)
+        dx, dw = ext.dw3x3_bwd(x, w32,
+                               _layout_contiguous(dy, ctx.nhwc),
+                               ctx.stride, need_dx, need_dw, ctx.nhwc)
         if need_dw:
             dw = dw.to(ctx.wdtype)
         return (dx if need_dx else None), (dw if need_dw else None), None
