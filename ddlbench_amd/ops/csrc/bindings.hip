@@ -7,7 +7,9 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
+#include <algorithm>
 #include <stdexcept>
+#include <unordered_map>
 #include <vector>
 
 // ---- launcher declarations (defined in the .hip kernel files) ---------
@@ -23,7 +25,7 @@ template <typename T>
 void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int,
                      hipStream_t);
 void set_bn_variant(int v);
-void launch_bn_finalize(const double*, float*, float*, float*, float*,
+void launch_bn_finalize(double*, float*, float*, float*, float*,
                         int64_t, double, float, float, hipStream_t);
 template <typename T>
 void launch_bn_apply(const T*, const T*, T*, const float*, const float*,
@@ -33,7 +35,7 @@ template <typename T>
 void launch_bn_bwd_reduce(const T*, const T*, const T*, const float*,
                           const float*, double*, int64_t, int64_t, int64_t,
                           int, int, hipStream_t);
-void launch_bn_bwd_finalize(const double*, const float*, const float*,
+void launch_bn_bwd_finalize(double*, const float*, const float*,
                             float*, float*, float*, int64_t, double, int,
                             hipStream_t);
 template <typename T>
@@ -88,6 +90,21 @@ bool is_bf16(const torch::Tensor& t) {
 template <typename T>
 T* dptr(const torch::Tensor& t) {
   return reinterpret_cast<T*>(t.data_ptr());
+}
+
+// Per-device BN reduction workspace. The finalize kernels write zeros
+// back after consuming, so after the first allocation no per-call fill
+// kernel runs (stream ordering serializes stats -> finalize -> reuse).
+torch::Tensor bn_sums_workspace(const torch::Tensor& like, int64_t C) {
+  static std::unordered_map<int, torch::Tensor> ws;
+  const int dev = (int)like.get_device();
+  auto it = ws.find(dev);
+  if (it == ws.end() || it->second.numel() < 2 * C) {
+    const int64_t cap = std::max<int64_t>(C, 2048);
+    ws[dev] = torch::zeros({2 * cap}, like.options().dtype(at::kDouble));
+    it = ws.find(dev);
+  }
+  return it->second;
 }
 
 }  // namespace
@@ -153,7 +170,7 @@ std::vector<torch::Tensor> bn_act_fwd(
   torch::Tensor mean = torch::empty({C}, fopt);
   torch::Tensor invstd = torch::empty({C}, fopt);
   if (training) {
-    torch::Tensor sums = torch::zeros({2, C}, x.options().dtype(at::kDouble));
+    torch::Tensor sums = bn_sums_workspace(x, C);
     if (is_bf16(x))
       launch_bn_stats<__hip_bfloat16>(dptr<__hip_bfloat16>(x),
                                       dptr<double>(sums), N, C, HW, nhwc, s);
@@ -198,7 +215,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto s = cur_stream();
   auto fopt = x.options().dtype(at::kFloat);
-  torch::Tensor sums = torch::zeros({2, C}, x.options().dtype(at::kDouble));
+  torch::Tensor sums = bn_sums_workspace(x, C);
   torch::Tensor dgamma = torch::empty({C}, fopt);
   torch::Tensor dbeta = torch::empty({C}, fopt);
   torch::Tensor k = torch::empty({3, C}, fopt);

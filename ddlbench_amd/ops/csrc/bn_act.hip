@@ -225,7 +225,7 @@ __global__ void bn_stats_nhwc_vec_f32_kernel(const float* __restrict__ x,
 }
 
 // ---- finalize: mean/invstd + running-stat update ----------------------
-__global__ void bn_finalize_kernel(const double* __restrict__ sums,
+__global__ void bn_finalize_kernel(double* __restrict__ sums,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
@@ -237,6 +237,8 @@ __global__ void bn_finalize_kernel(const double* __restrict__ sums,
   const double m = sums[c] / count;
   double var = sums[C + c] / count - m * m;
   var = var < 0.0 ? 0.0 : var;
+  sums[c] = 0.0;      // leave the workspace zeroed for the next caller
+  sums[C + c] = 0.0;
   mean[c] = (float)m;
   invstd[c] = (float)rsqrt(var + (double)eps);
   if (running_mean != nullptr) {
@@ -653,7 +655,7 @@ __global__ void bn_bwd_reduce_nhwc_vec_kernel(
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
+__global__ void bn_bwd_finalize_kernel(double* __restrict__ sums,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ invstd,
                                        float* __restrict__ dgamma,
@@ -664,6 +666,8 @@ __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
   const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const double sdy = sums[c], sdyx = sums[C + c];
+  sums[c] = 0.0;
+  sums[C + c] = 0.0;
   dgamma[c] = (float)sdyx;
   dbeta[c] = (float)sdy;
   k[c] = gamma[c] * invstd[c];                       // k1
@@ -736,7 +740,7 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   HIP_CHECK_LAST();
 }
 
-void launch_bn_finalize(const double* sums, float* mean, float* invstd,
+void launch_bn_finalize(double* sums, float* mean, float* invstd,
                         float* rm, float* rv, int64_t C, double count,
                         float eps, float momentum, hipStream_t stream) {
   const int block = 256;
@@ -859,7 +863,7 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   HIP_CHECK_LAST();
 }
 
-void launch_bn_bwd_finalize(const double* sums, const float* gamma,
+void launch_bn_bwd_finalize(double* sums, const float* gamma,
                             const float* invstd, float* dgamma, float* dbeta,
                             float* k, int64_t C, double count, int training,
                             hipStream_t stream) {

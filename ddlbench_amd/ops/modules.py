@@ -49,6 +49,7 @@ class BNAct(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
+        self._nbt = 0
 
     def _apply(self, fn, recurse=True):
         # keep BN params/stats fp32 under model.to(bf16): re-cast after
@@ -68,11 +69,19 @@ class BNAct(nn.Module):
     def forward(self, x: torch.Tensor,
                 res: Optional[torch.Tensor] = None) -> torch.Tensor:
         if self.training:
-            self.num_batches_tracked += 1
+            # python-side counter; synced to the buffer on state_dict()
+            # (a per-step GPU add showed up as 51 launches/step)
+            self._nbt += 1
         return NF.bn_act(x, self.weight, self.bias, self.running_mean,
                          self.running_var, self.training, self.momentum,
                          self.eps, self.act, res,
                          backend=_DEFAULT_BACKEND)
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        if self._nbt:
+            self.num_batches_tracked += self._nbt
+            self._nbt = 0
+        super()._save_to_state_dict(destination, prefix, keep_vars)
 
     def extra_repr(self) -> str:
         return f"{self.num_features}, act={self.act}"
