@@ -31,6 +31,8 @@ def run_single(cfg: BenchConfig) -> dict:
     model = build_model(cfg.dataset, cfg.arch).to(device)
     if compute_dtype(cfg) != torch.float32:
         model = model.to(compute_dtype(cfg))
+    if cfg.channels_last and device.type == "cuda":
+        model = model.to(memory_format=torch.channels_last)
     optimizer = make_optimizer(cfg, model)
     start_epoch = 1
     if cfg.resume and cfg.checkpoint_dir:
@@ -67,6 +69,8 @@ def run_ddp(cfg: BenchConfig) -> dict:
     model = build_model(cfg.dataset, cfg.arch).to(device)
     if compute_dtype(cfg) != torch.float32:
         model = model.to(compute_dtype(cfg))
+    if cfg.channels_last and device.type == "cuda":
+        model = model.to(memory_format=torch.channels_last)
     dp = BucketedDataParallel(model)
     # reference scales LR by world size (mnist_horovod.py:226)
     optimizer = make_optimizer(cfg, model, lr_scale=env.world_size)

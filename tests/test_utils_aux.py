@@ -95,3 +95,21 @@ def test_fused_adamw_matches_torch():
             o.step()
     for pa, pb in zip(a.parameters(), b.parameters()):
         torch.testing.assert_close(pa, pb, rtol=1e-5, atol=1e-6)
+
+
+def test_conv_mfma_module_cpu_fallback():
+    """Conv2dMFMA uses the library conv on CPU (same math)."""
+    import torch.nn as nn
+    from ddlbench_amd.ops.conv import Conv2dMFMA, convert_convs, mfma_eligible
+    torch.manual_seed(0)
+    conv = nn.Conv2d(16, 32, 3, stride=1, padding=1, bias=False)
+    assert mfma_eligible(conv)
+    wrapped = Conv2dMFMA(conv)
+    x = torch.randn(2, 16, 8, 8)
+    torch.testing.assert_close(wrapped(x), conv(x))
+    # converter counts + skips the C=3 stem
+    model = nn.Sequential(nn.Conv2d(3, 16, 3, bias=False),
+                          nn.Conv2d(16, 16, 3, bias=False))
+    assert convert_convs(model) == 1
+    assert isinstance(model[0], nn.Conv2d)
+    assert isinstance(model[1], Conv2dMFMA)
