@@ -123,3 +123,75 @@ def _worker_runner(rank, world, port):
 @pytest.mark.timeout(600)
 def test_full_1f1b_runner_two_stages(free_port):
     mp.spawn(_worker_runner, args=(2, free_port), nprocs=2, join=True)
+
+
+# ---------------------------------------- hybrid: replicated stage 0
+def _worker_hybrid(rank, world, port):
+    _env(rank, world, port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from ddlbench_amd.ops.sgd import FusedSGD
+    from ddlbench_amd.parallel import BucketedDataParallel
+    from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
+    from ddlbench_amd.parallel.pipeline.runtime import (StagePlan,
+                                                        StageRuntime)
+    from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
+
+    torch.manual_seed(0)
+    full = torch.nn.Sequential(
+        torch.nn.Linear(6, 16), torch.nn.Tanh(), torch.nn.Linear(16, 3))
+    plan = StagePlan(replicas=[2, 1])  # ranks 0,1 -> stage0; rank 2 -> stage1
+    assert plan.edges() == [(0, 2), (1, 2)]
+    assert plan.num_warmup(0) == 0 and plan.num_warmup(1) == 0
+    tr = PipelineTransport(plan.edges(), backend="gloo")
+    stage, replica = plan.stage_of_rank(rank)
+    mod = (torch.nn.Sequential(*list(full)[:2]) if stage == 0
+           else torch.nn.Sequential(*list(full)[2:]))
+    dp = None
+    g = dist.new_group(plan.stage_ranks(0))
+    if stage == 0:
+        dp = BucketedDataParallel(mod, process_group=g, bucket_mb=1)
+    B = 4
+    rt = StageRuntime(plan, rank, mod, tr,
+                      in_shape=torch.Size([B, 16]) if stage else None,
+                      out_shape=torch.Size([B, 16]) if stage == 0
+                      else torch.Size([B, 3]),
+                      device=torch.device("cpu"), dtype=torch.float32,
+                      loss_fn=torch.nn.functional.cross_entropy,
+                      dp_wrapper=dp)
+    opt = VersionedOptimizer(
+        FusedSGD(mod.parameters(), lr=0.05, momentum=0.9,
+                 backend="torch"), versioned=False)
+
+    gen = torch.Generator().manual_seed(11)
+    N = 8
+    xs = [torch.randn(B, 6, generator=gen) for _ in range(N)]
+    ys = [torch.randint(3, (B,), generator=gen) for _ in range(N)]
+    mbs = rt.my_minibatches(N)
+    assert (len(mbs) == 4) if stage == 0 else (len(mbs) == 8)
+    for mb in mbs:
+        rt.run_forward(mb, lambda i: xs[i], lambda i: ys[i], training=True)
+        if dp is not None:
+            dp.zero_grad_buckets()
+        else:
+            opt.zero_grad(set_to_none=False)
+        rt.run_backward()
+        if dp is not None:
+            dp.finalize_backward()
+        opt.step()
+    # stage-0 replicas must hold identical weights after DP averaging
+    if stage == 0:
+        for p in mod.parameters():
+            ref = p.data.clone()
+            dist.broadcast(ref, src=0, group=g)
+            torch.testing.assert_close(ref, p.data)
+    assert rt.stats.fwd_count == len(mbs)
+    assert rt.stats.bwd_count == len(mbs)
+    if stage == 0:
+        assert rt.stats.send_bytes == len(mbs) * B * 16 * 4
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_hybrid_replicated_stage(free_port):
+    mp.spawn(_worker_hybrid, args=(3, free_port), nprocs=3, join=True)
