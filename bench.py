@@ -42,6 +42,12 @@ def parse_args():
                    help="auto|cuda|cpu (cpu only for plumbing tests)")
     p.add_argument("--kernel-backend", default="auto",
                    choices=["auto", "native", "torch"])
+    p.add_argument("--strategy", default="ddp", choices=["ddp", "gpipe"],
+                   help="ddp: one rank per GPU over RCCL (the driver "
+                        "contract); gpipe: single-process micro-batch "
+                        "pipeline across all visible GPUs")
+    p.add_argument("--microbatches", type=int, default=8,
+                   help="gpipe chunk count")
     p.add_argument("--conv", default="miopen", choices=["miopen", "mfma"],
                    help="conv backend: library (MIOpen) or the in-tree "
                         "MFMA implicit-GEMM kernels")
@@ -78,11 +84,25 @@ def main():
     channels_last = (args.memory_format == "channels_last"
                      and device.type == "cuda")
     torch.manual_seed(1234 + env.rank)
-    model = build_model(cfg.dataset, cfg.arch).to(device)
-    if dtype != torch.float32:
-        model = model.to(dtype)
-    if channels_last:
-        model = model.to(memory_format=torch.channels_last)
+    if args.strategy == "gpipe":
+        from ddlbench_amd.models import build_sequential
+        from ddlbench_amd.parallel.pipeline.gpipe import build_gpipe
+        assert world == 1, "gpipe strategy is single-process multi-device"
+        seq = build_sequential(cfg.dataset, cfg.arch)
+        if dtype != torch.float32:
+            seq = seq.to(dtype)
+        sample = torch.randn(max(args.batch // args.microbatches, 1),
+                             3, 224, 224, dtype=dtype)
+        import dataclasses
+        gcfg = dataclasses.replace(cfg, microbatches=args.microbatches)
+        model = build_gpipe(gcfg, seq, sample.to(device))
+        device = model.out_device
+    else:
+        model = build_model(cfg.dataset, cfg.arch).to(device)
+        if dtype != torch.float32:
+            model = model.to(dtype)
+        if channels_last:
+            model = model.to(memory_format=torch.channels_last)
     if args.conv == "mfma" and device.type == "cuda":
         from ddlbench_amd.ops.conv import convert_convs
         n_conv = convert_convs(model, dtype)
@@ -158,7 +178,9 @@ def main():
                 "model": args.model,
                 "global_batch": args.batch * world,
                 "input": "3x224x224",
-                "parallelism": f"dp{world}",
+                "parallelism": (f"dp{world}" if args.strategy == "ddp"
+                                else f"gpipe{torch.cuda.device_count() or 1}"
+                                     f"x{args.microbatches}mb"),
             },
         }), flush=True)
 
