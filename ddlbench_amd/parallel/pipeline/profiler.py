@@ -59,12 +59,11 @@ def profile_sequential(seq: nn.Sequential, sample: torch.Tensor,
             if record:
                 fwd_t[i] += time.perf_counter() - t0
                 act_bytes[i] = x.numel() * x.element_size()
-        # backward: per layer, using the saved graph of that layer only
+        # backward: per layer. The timed forward above used detached
+        # inputs, so each layer's graph is isolated; for non-final
+        # layers re-run the forward (untimed) to get a fresh graph.
         dy = torch.ones_like(x)
         for i in range(n - 1, -1, -1):
-            out = x if i == n - 1 else None
-            # re-run layer i on its stored input to get a fresh graph
-            # (outputs of the fwd pass above are consumed layer by layer)
             y_i = seq[i](inputs[i]) if i != n - 1 else x
             grad_out = dy if i == n - 1 else torch.ones_like(y_i)
             _sync(device)

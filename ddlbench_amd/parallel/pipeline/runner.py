@@ -195,6 +195,7 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
             loss, _ = rt.run_forward(mbs[k], inp, tgt, training=True)
             if loss is not None:
                 loss_vals.append(loss.detach())
+        window_start, window_n = time.perf_counter(), 0
         for k in range(len(mbs)):
             if warmup + k < len(mbs):
                 loss, _ = rt.run_forward(mbs[warmup + k], inp, tgt,
@@ -209,6 +210,19 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
             if dp is not None:
                 dp.finalize_backward()
             opt.step()
+            window_n += 1
+            if (cfg.log_interval and is_output
+                    and (k + 1) % cfg.log_interval == 0):
+                sync()
+                now = time.perf_counter()
+                from ddlbench_amd.utils import gpu_memory_gb
+                sps_w = (window_n * B * plan.replicas[-1]
+                         / (now - window_start))
+                a, r, t = gpu_memory_gb(device)
+                log.train_step(epoch, cfg.epochs,
+                               int(100 * (k + 1) / len(mbs)), sps_w,
+                               a, r, t)
+                window_start, window_n = time.perf_counter(), 0
         sync()
         if dist.is_initialized():
             dist.barrier()
