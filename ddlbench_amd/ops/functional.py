@@ -124,6 +124,8 @@ class _FusedDWConv3x3(torch.autograd.Function):
         ctx.save_for_backward(x, w32)
         ctx.stride = stride
         ctx.wdtype = weight.dtype
+        ctx.w_cl = weight.is_contiguous(memory_format=torch.channels_last) \
+            and not weight.is_contiguous()
         ctx.nhwc = nhwc
         return y
 
@@ -137,6 +139,8 @@ class _FusedDWConv3x3(torch.autograd.Function):
                                ctx.stride, need_dx, need_dw, ctx.nhwc)
         if need_dw:
             dw = dw.to(ctx.wdtype)
+            if ctx.w_cl:  # match a channels_last weight's layout
+                dw = dw.contiguous(memory_format=torch.channels_last)
         return (dx if need_dx else None), (dw if need_dw else None), None
 
 

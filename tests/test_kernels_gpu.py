@@ -263,6 +263,53 @@ def test_conv_mfma_resnet_block_trains():
     assert torch.isfinite(loss)
 
 
+def test_mobilenet_channels_last_train_step():
+    """MobileNetV2 bf16 channels_last: full fwd+bwd+fused-SGD step (the
+    depthwise weight-grad layout must match the converted param)."""
+    from ddlbench_amd.models import build_model
+    from ddlbench_amd.ops import functional as NF
+    from ddlbench_amd.ops.sgd import FusedSGD
+    torch.manual_seed(0)
+    dev = _dev()
+    m = build_model("cifar10", "mobilenetv2").to(dev).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    opt = FusedSGD(m.parameters(), lr=0.01, momentum=0.9, backend="native")
+    x = torch.randn(16, 3, 32, 32, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(10, (16,), device=dev)
+    loss = NF.cross_entropy(m(x), y, backend="native")
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
+
+
+@pytest.mark.parametrize("stride", [1, 2])
+def test_depthwise_nhwc_matches_nchw(stride):
+    """NHWC depthwise kernels vs the fp32 torch reference."""
+    from ddlbench_amd.ops import functional as NF
+    torch.manual_seed(0)
+    dev = _dev()
+    N, C, H, W = 4, 32, 14, 14
+    x1 = torch.randn(N, C, H, W, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    w1 = torch.randn(C, 1, 3, 3, device=dev, dtype=torch.bfloat16,
+                     requires_grad=True)
+    y1 = NF.depthwise_conv3x3(x1, w1, stride, backend="native")
+    assert y1.is_contiguous(memory_format=torch.channels_last)
+    dy = torch.randn_like(y1)
+    y1.backward(dy)
+    x2 = x1.detach().float().contiguous().requires_grad_(True)
+    w2 = w1.detach().float().clone().requires_grad_(True)
+    y2 = torch.nn.functional.conv2d(x2, w2, None, stride, 1, 1, groups=C)
+    y2.backward(dy.float().contiguous())
+    torch.testing.assert_close(y1.float().contiguous(), y2, rtol=3e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(x1.grad.float().contiguous(), x2.grad,
+                               rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=3e-2,
+                               atol=0.3)
+
+
 # ------------------------------------------------------------ fused SGD
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_fused_sgd_matches_torch(dtype):
