@@ -68,6 +68,8 @@ class BenchConfig:
     kernel_backend: str = "auto"      # auto | native | torch (ops dispatch)
     checkpoint_dir: str = ""          # per-stage checkpoints when set
     resume: bool = False              # load stage checkpoints at start
+    no_input_pipelining: bool = False  # pipedream: pure model parallel
+    straight_pipeline: bool = False    # pipedream: no stage replication
 
     def __post_init__(self) -> None:
         if self.dataset not in DATASET_SHAPES:

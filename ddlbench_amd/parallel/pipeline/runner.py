@@ -41,8 +41,8 @@ from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
 from ddlbench_amd.utils import AverageMeter, BenchLogger, accuracy
 
 
-def _make_plan(cfg: BenchConfig, seq, device, world: int,
-               straight: bool = False):
+def _make_plan(cfg: BenchConfig, seq, device, world: int):
+    straight = cfg.straight_pipeline
     """Rank 0 profiles + partitions; everyone gets the same plan."""
     rank = dist.get_rank() if dist.is_initialized() else 0
     payload = [None]
@@ -127,7 +127,8 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
         FusedSGD(stage_mod.parameters(), lr=cfg.lr, momentum=cfg.momentum,
                  weight_decay=cfg.weight_decay,
                  backend=cfg.kernel_backend),
-        versioned=plan.num_warmup(stage) > 0)
+        versioned=(not cfg.no_input_pipelining
+                   and plan.num_warmup(stage) > 0))
 
     # deterministic synthetic streams: stage 0 reads inputs, last stage
     # reads the matching targets (no label transport needed)
@@ -185,7 +186,10 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
         stage_mod.train()
         n_mb, inp, tgt = make_providers(train_ds, epoch)
         mbs = rt.my_minibatches(n_mb)
-        warmup = min(plan.num_warmup(stage), len(mbs))
+        # --no_input_pipelining: pure model parallel, one minibatch in
+        # flight (reference main_with_runtime.py:66-67, 233-235)
+        warmup = (0 if cfg.no_input_pipelining
+                  else min(plan.num_warmup(stage), len(mbs)))
         loss_vals = []  # detached tensors; .item() deferred to epoch end
         if dist.is_initialized():
             dist.barrier()
