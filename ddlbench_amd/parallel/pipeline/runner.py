@@ -42,8 +42,8 @@ from ddlbench_amd.utils import AverageMeter, BenchLogger, accuracy
 
 
 def _make_plan(cfg: BenchConfig, seq, device, world: int):
-    straight = cfg.straight_pipeline
     """Rank 0 profiles + partitions; everyone gets the same plan."""
+    straight = cfg.straight_pipeline
     rank = dist.get_rank() if dist.is_initialized() else 0
     payload = [None]
     if rank == 0:
