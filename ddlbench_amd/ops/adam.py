@@ -10,6 +10,7 @@ import torch
 from torch.optim import Optimizer
 
 from ddlbench_amd import ops as _ops
+from ddlbench_amd.ops.sgd import _same_dense_layout
 
 
 class FusedAdam(Optimizer):
@@ -108,7 +109,11 @@ class FusedAdam(Optimizer):
                 }
                 self._cache[key] = cached
             for p in ps:
-                assert p.grad.stride() == p.stride()
+                if not _same_dense_layout(p, p.grad):
+                    p.grad = p.grad.contiguous() if p.is_contiguous() \
+                        else p.grad.contiguous(
+                            memory_format=torch.channels_last)
+                assert _same_dense_layout(p, p.grad)
             ptr_grads = torch.tensor([p.grad.data_ptr() for p in ps],
                                      dtype=torch.int64).to(
                                          dev, non_blocking=True)

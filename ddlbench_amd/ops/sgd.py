@@ -13,6 +13,18 @@ from torch.optim import Optimizer
 from ddlbench_amd import ops as _ops
 
 
+def _same_dense_layout(p, g):
+    """True when grad and param walk memory in the same element order.
+    Stride values on size-1 dims are arbitrary (a (K,C,1,1) weight is
+    simultaneously 'contiguous' and 'channels_last'), so compare strides
+    only where the dim extent is > 1."""
+    if p.shape != g.shape:
+        return False
+    return all(sz <= 1 or ps == gs
+               for sz, ps, gs in zip(p.shape, p.stride(), g.stride()))
+
+
+
 class FusedSGD(Optimizer):
     def __init__(self, params, lr: float, momentum: float = 0.0,
                  weight_decay: float = 0.0, backend: str = "auto"):
@@ -102,7 +114,11 @@ class FusedSGD(Optimizer):
                 # the fused kernel walks param/grad as flat buffers in
                 # storage order — any dense layout works (NCHW or
                 # channels_last) as long as grad strides match param's
-                assert p.grad.stride() == p.stride(), \
+                if not _same_dense_layout(p, p.grad):
+                    p.grad = p.grad.contiguous() if p.is_contiguous() \
+                        else p.grad.contiguous(
+                            memory_format=torch.channels_last)
+                assert _same_dense_layout(p, p.grad), \
                     "FusedSGD needs grads with the param's layout"
             # grads in persistent buckets (DP) keep their pointers —
             # re-upload the pointer array only when one moved
