@@ -100,21 +100,32 @@ DEV const bf16* b_chunk_addr(const ConvParams& p, long row, long kkg) {
 // where the DATA stored is logical cg = cg' ^ (row & 7). ds_read applies
 // the same XOR.
 
-template <bool DGRAD>
+// Tile geometry is templated so small output-channel counts get a
+// narrow N tile instead of half-empty MFMA work: 4 waves arranged
+// (WR x WC) each owning a (BM_/WR x BN_/WC) sub-tile.
+template <bool DGRAD, int BM_, int BN_, int WR, int WC>
 __global__ __launch_bounds__(THREADS, 2)
 void conv_igemm_kernel(ConvParams p) {
+  constexpr int WM = BM_ / WR;        // wave tile M
+  constexpr int WN = BN_ / WC;        // wave tile N
+  constexpr int MF = WM / 16;         // M fragments per wave
+  constexpr int NF = WN / 16;         // N fragments per wave
+  constexpr int CA = BM_ * BK / 8 / THREADS;  // A chunks per thread
+  constexpr int CB = BN_ * BK / 8 / THREADS;  // B chunks per thread
+  static_assert(WR * WC == 4, "4 waves");
+  static_assert(CA >= 1 && CB >= 1, "tile vs thread count");
+
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* lds = reinterpret_cast<bf16*>(smem);
-  // buffers: [buf][AB][128*64]
   auto lds_tile = [&](int buf, int ab) {
-    return lds + ((buf * 2 + ab) * BM * BK);
+    return lds + buf * (BM_ + BN_) * BK + (ab ? BM_ * BK : 0);
   };
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
-  const int nbn = (int)((p.Nd + BN - 1) / BN);
+  const int nbn = (int)((p.Nd + BN_ - 1) / BN_);
   int block = blockIdx.x;
   // XCD-aware swizzle: contiguous chunks per XCD (bijective form)
   {
@@ -124,20 +135,23 @@ void conv_igemm_kernel(ConvParams p) {
     block = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
             + idx;
   }
-  const long bm = (long)(block / nbn) * BM;
-  const long bn = (long)(block % nbn) * BN;
+  const long bm = (long)(block / nbn) * BM_;
+  const long bn = (long)(block % nbn) * BN_;
 
   const int nsteps = (int)((p.Kd + BK - 1) / BK);
 
-  // ---- staging: each thread owns 4 A-chunks and 4 B-chunks ----------
-  // slot ca = l*256 + tid; row = ca>>3, cg_store = ca&7,
-  // data(cg) = cg_store ^ (row&7)
-  int a_row[4], a_cg[4], b_row[4], b_cg[4];
+  // ---- staging slots: slot ca -> (row = ca>>3, stored cg = ca&7),
+  // holding source logical cg = (ca&7) ^ (row&7)
+  int a_row[CA], a_cg[CA], b_row[CB], b_cg[CB];
 #pragma unroll
-  for (int l = 0; l < 4; ++l) {
+  for (int l = 0; l < CA; ++l) {
     const int ca = l * THREADS + tid;
     a_row[l] = ca >> 3;
     a_cg[l] = (ca & 7) ^ (a_row[l] & 7);
+  }
+#pragma unroll
+  for (int l = 0; l < CB; ++l) {
+    const int ca = l * THREADS + tid;
     b_row[l] = ca >> 3;
     b_cg[l] = (ca & 7) ^ (b_row[l] & 7);
   }
@@ -147,7 +161,7 @@ void conv_igemm_kernel(ConvParams p) {
     bf16* la = lds_tile(buf, 0);
     bf16* lb = lds_tile(buf, 1);
 #pragma unroll
-    for (int l = 0; l < 4; ++l) {
+    for (int l = 0; l < CA; ++l) {
       const long kkga = kk0 + a_cg[l] * 8;
       const bf16* src = a_chunk_addr<DGRAD>(p, bm + a_row[l],
                                             kkga < p.Kd ? kkga : 0);
@@ -158,7 +172,7 @@ void conv_igemm_kernel(ConvParams p) {
               (l * THREADS + tid) * 8), 16, 0, 0);
     }
 #pragma unroll
-    for (int l = 0; l < 4; ++l) {
+    for (int l = 0; l < CB; ++l) {
       const bf16* src = b_chunk_addr(p, bn + b_row[l],
                                      kk0 + b_cg[l] * 8);
       __builtin_amdgcn_global_load_lds(
@@ -169,23 +183,21 @@ void conv_igemm_kernel(ConvParams p) {
   };
 
   // ---- fragment read offsets (XOR re-applied) -----------------------
-  // A frag (16x16x32): lane reads row m0 + (lane&15), elems
-  // ks*32 + (lane>>4)*8 .. +7  ->  cg = ks*4 + (lane>>4)
   const int fr = lane & 15;         // row-in-frag
   const int fk = lane >> 4;         // k-subgroup
-  const int wm = (wid >> 1) * 64;   // wave M offset
-  const int wn = (wid & 1) * 64;    // wave N offset
+  const int wm = (wid / WC) * WM;   // wave M offset
+  const int wn = (wid % WC) * WN;   // wave N offset
 
   auto frag_ptr = [&](bf16* tile, int row, int ks) -> const bf16x8* {
     const int cg = (ks * 4 + fk) ^ (row & 7);
     return reinterpret_cast<const bf16x8*>(tile + row * BK + cg * 8);
   };
 
-  f32x4 acc[4][4];
+  f32x4 acc[MF][NF];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < MF; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
+    for (int j = 0; j < NF; ++j)
       acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   stage(0, 0);
@@ -199,18 +211,18 @@ void conv_igemm_kernel(ConvParams p) {
     bf16* lb = lds_tile(cur, 1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[MF], bfr[NF];
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < MF; ++mf)
         af[mf] = *frag_ptr(la, wm + mf * 16 + fr, ks);
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
+      for (int nf = 0; nf < NF; ++nf)
         bfr[nf] = *frag_ptr(lb, wn + nf * 16 + fr, ks);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+      for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
+        for (int nf = 0; nf < NF; ++nf)
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -222,13 +234,13 @@ void conv_igemm_kernel(ConvParams p) {
 
   // ---- epilogue: D[row=(lane>>4)*4+reg][col=lane&15] per fragment ---
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const long row = bm + wm + mf * 16 + fk * 4 + reg;
       if (row >= p.M) continue;
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NF; ++nf) {
         const long col = bn + wn + nf * 16 + fr;
         if (col < p.Nd)
           p.out[row * p.Nd + col] = from_f32<bf16>(acc[mf][nf][reg]);
@@ -258,15 +270,27 @@ void launch_conv_igemm(const void* a, const void* b, void* out,
     p.Nd = Cin;
     p.Kd = (long)R * S * K;
   }
-  const long nbm = (p.M + BM - 1) / BM;
-  const long nbn = (p.Nd + BN - 1) / BN;
-  const size_t lds_bytes = 2 * 2 * BM * BK * sizeof(bf16);
-  const dim3 grid((unsigned)(nbm * nbn));
-  if (!dgrad)
-    hipLaunchKernelGGL((conv_igemm_kernel<false>), grid, dim3(THREADS),
-                       lds_bytes, stream, p);
-  else
-    hipLaunchKernelGGL((conv_igemm_kernel<true>), grid, dim3(THREADS),
-                       lds_bytes, stream, p);
+
+#define LAUNCH(DG, BM_, BN_, WR, WC)                                        \
+  do {                                                                      \
+    const long nbm = (p.M + (BM_) - 1) / (BM_);                             \
+    const long nbn = (p.Nd + (BN_) - 1) / (BN_);                            \
+    const size_t lds_bytes = 2 * ((BM_) + (BN_)) * BK * sizeof(bf16);       \
+    hipLaunchKernelGGL((conv_igemm_kernel<DG, BM_, BN_, WR, WC>),           \
+                       dim3((unsigned)(nbm * nbn)), dim3(THREADS),          \
+                       lds_bytes, stream, p);                               \
+  } while (0)
+
+  // narrow-N tiles when the output-channel dim can't fill 128 columns
+  if (!dgrad) {
+    if (p.Nd <= 64 && p.M >= 256) LAUNCH(false, 256, 64, 4, 1);
+    else if (p.Nd <= 64) LAUNCH(false, 128, 64, 4, 1);
+    else LAUNCH(false, 128, 128, 2, 2);
+  } else {
+    if (p.Nd <= 64 && p.M >= 256) LAUNCH(true, 256, 64, 4, 1);
+    else if (p.Nd <= 64) LAUNCH(true, 128, 64, 4, 1);
+    else LAUNCH(true, 128, 128, 2, 2);
+  }
+#undef LAUNCH
   HIP_CHECK_LAST();
 }
