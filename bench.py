@@ -36,6 +36,9 @@ def parse_args():
     p.add_argument("--batch", type=int, default=256,
                    help="per-GPU batch size")
     p.add_argument("--model", default="resnet50")
+    p.add_argument("--dataset", default="imagenet",
+                   choices=["imagenet", "cifar10", "mnist", "highres"],
+                   help="input shape / class count")
     p.add_argument("--dtype", default="bfloat16",
                    choices=["bfloat16", "float32"])
     p.add_argument("--device", default="auto",
@@ -71,7 +74,7 @@ def main():
 
     env = init_distributed()
     world = env.world_size
-    cfg = BenchConfig(dataset="imagenet", arch=args.model, strategy="ddp",
+    cfg = BenchConfig(dataset=args.dataset, arch=args.model, strategy="ddp",
                       batch_size=args.batch, dtype=args.dtype,
                       kernel_backend=args.kernel_backend, num_workers=0)
     set_default_backend(cfg.kernel_backend)
@@ -91,8 +94,8 @@ def main():
         seq = build_sequential(cfg.dataset, cfg.arch)
         if dtype != torch.float32:
             seq = seq.to(dtype)
-        sample = torch.randn(max(args.batch // args.microbatches, 1),
-                             3, 224, 224, dtype=dtype)
+        sample = torch.randn((max(args.batch // args.microbatches, 1),)
+                             + tuple(cfg.shape), dtype=dtype)
         import dataclasses
         gcfg = dataclasses.replace(cfg, microbatches=args.microbatches)
         model = build_gpipe(gcfg, seq, sample.to(device))
@@ -119,8 +122,8 @@ def main():
     pool = []
     g = torch.Generator().manual_seed(7 + env.rank)
     for _ in range(n_pool):
-        x = torch.randn(args.batch, 3, 224, 224, generator=g)
-        y = torch.randint(1000, (args.batch,), generator=g)
+        x = torch.randn((args.batch,) + tuple(cfg.shape), generator=g)
+        y = torch.randint(cfg.num_classes, (args.batch,), generator=g)
         x = x.to(device, dtype=dtype)
         if channels_last:
             x = x.contiguous(memory_format=torch.channels_last)
@@ -177,7 +180,7 @@ def main():
             "config": {
                 "model": args.model,
                 "global_batch": args.batch * world,
-                "input": "3x224x224",
+                "input": "x".join(str(d) for d in cfg.shape),
                 "parallelism": (f"dp{world}" if args.strategy == "ddp"
                                 else f"gpipe{torch.cuda.device_count() or 1}"
                                      f"x{args.microbatches}mb"),
