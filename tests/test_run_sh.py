@@ -1,0 +1,21 @@
+"""run.sh front-end smoke (CPU)."""
+import subprocess
+
+
+def test_run_sh_pytorch_cpu(tmp_path):
+    out = subprocess.run(
+        ["bash", "run/run.sh", "-b", "mnist", "-f", "pytorch",
+         "-m", "resnet18", "-e", "1", "-B", "32", "-p", "0"],
+        capture_output=True, text=True, timeout=900,
+        env={"PATH": "/usr/bin:/bin:/usr/local/bin",
+             "DDLB_DEVICE": "cpu", "HOME": str(tmp_path)})
+    # the entrypoint runs on whatever device exists; here CPU
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "valid accuracy:" in out.stdout
+
+
+def test_run_sh_rejects_bad_framework():
+    out = subprocess.run(
+        ["bash", "run/run.sh", "-f", "nope"],
+        capture_output=True, text=True, timeout=60)
+    assert out.returncode != 0
