@@ -333,9 +333,15 @@ torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
   const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
   const int R = w_perm.size(1), S = w_perm.size(2);
   TORCH_CHECK(K % 8 == 0, "conv_igemm dgrad needs K % 8 == 0");
-  auto dx = torch::empty({N, C, H, W},
-                         dy.options().memory_format(
-                             at::MemoryFormat::ChannelsLast));
+  // stride-2 parity classes with R==1 or S==1 have pixel classes no tap
+  // reaches — those dx entries must be zeros, so zero-init in that case
+  auto dx = (stride == 2 && (R == 1 || S == 1))
+      ? torch::zeros({N, C, H, W},
+                     dy.options().memory_format(
+                         at::MemoryFormat::ChannelsLast))
+      : torch::empty({N, C, H, W},
+                     dy.options().memory_format(
+                         at::MemoryFormat::ChannelsLast));
   auto zero = torch::zeros({16}, dy.options());
   launch_conv_igemm(dy.data_ptr(), w_perm.data_ptr(), dx.data_ptr(),
                     zero.data_ptr(), (int)N, (int)H, (int)W, (int)C, K,

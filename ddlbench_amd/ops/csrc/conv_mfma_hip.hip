@@ -49,16 +49,22 @@ struct ConvParams {
   int K, OH, OW;      // output channels / spatial of the forward conv
   int R, S, stride, pad;
   long M, Nd, Kd;     // GEMM dims of THIS pass
+  // stride-2 dgrad parity class (MODE 2): output pixels with
+  // ih%2==cls_a, iw%2==cls_b; valid taps r = r0+2*i (nr of them),
+  // s = s0+2*j (ns of them); nh/nw = pixel counts of the class
+  int cls_a, cls_b, r0, s0, nr, ns, nh, nw;
 };
 
 // ---- chunk -> global address generators -------------------------------
 // A-chunk: logical (row=m in [0,BM), cg in [0,8)) of the current K-step.
 // Returns the 16-byte-aligned source for elements kk0+cg*8 .. +7.
 
-template <bool DGRAD>
+// MODE: 0 = FPROP, 1 = DGRAD (any stride, gather with validity tests),
+//       2 = DGRAD stride-2 parity class (all taps valid by construction)
+template <int MODE>
 DEV const bf16* a_chunk_addr(const ConvParams& p, long m, long kkg) {
   if (m >= p.M) return p.zero;
-  if (!DGRAD) {
+  if (MODE == 0) {
     // m -> (n, oh, ow); kkg -> (r, s, c0)
     const int ohw = p.OH * p.OW;
     const int n = (int)(m / ohw);
@@ -71,7 +77,7 @@ DEV const bf16* a_chunk_addr(const ConvParams& p, long m, long kkg) {
     const int iw = ow * p.stride + s - p.pad;
     if (ih < 0 || ih >= p.H || iw < 0 || iw >= p.W) return p.zero;
     return p.a + (((long)n * p.H + ih) * p.W + iw) * p.Cin + c0;
-  } else {
+  } else if (MODE == 1) {
     // m -> (n, ih, iw); kkg -> (r, s, k0); gather from dy
     const int hw = p.H * p.W;
     const int n = (int)(m / hw);
@@ -87,12 +93,35 @@ DEV const bf16* a_chunk_addr(const ConvParams& p, long m, long kkg) {
     const int oh = tih / p.stride, ow = tiw / p.stride;
     if (oh >= p.OH || ow >= p.OW) return p.zero;
     return p.a + (((long)n * p.OH + oh) * p.OW + ow) * p.K + k0;
+  } else {
+    // stride-2 class: m -> (n, ii, jj); ih = 2*ii+cls_a; only the class's
+    // valid (r, s) taps are enumerated, so no divisibility test
+    const int hw = p.nh * p.nw;
+    const int n = (int)(m / hw);
+    const int rem = (int)(m - (long)n * hw);
+    const int ii = rem / p.nw, jj = rem - (rem / p.nw) * p.nw;
+    const int ih = 2 * ii + p.cls_a, iw = 2 * jj + p.cls_b;
+    const int k0 = (int)(kkg % p.K);
+    const int rs = (int)(kkg / p.K);
+    const int ri = rs / p.ns, si = rs - (rs / p.ns) * p.ns;
+    const int r = p.r0 + 2 * ri, s = p.s0 + 2 * si;
+    const int oh = (ih + p.pad - r) >> 1;
+    const int ow = (iw + p.pad - s) >> 1;
+    if (oh < 0 || ow < 0 || oh >= p.OH || ow >= p.OW) return p.zero;
+    return p.a + (((long)n * p.OH + oh) * p.OW + ow) * p.K + k0;
   }
 }
 
+template <int MODE>
 DEV const bf16* b_chunk_addr(const ConvParams& p, long row, long kkg) {
   if (row >= p.Nd || kkg >= p.Kd) return p.zero;
-  return p.b + row * p.Kd + kkg;
+  if (MODE != 2) return p.b + row * p.Kd + kkg;
+  // class subset of the (C,R,S,K) weight image
+  const int k0 = (int)(kkg % p.K);
+  const int rs = (int)(kkg / p.K);
+  const int ri = rs / p.ns, si = rs - (rs / p.ns) * p.ns;
+  const int r = p.r0 + 2 * ri, s = p.s0 + 2 * si;
+  return p.b + (row * p.R + r) * (long)p.S * p.K + (long)s * p.K + k0;
 }
 
 // ---- the kernel -------------------------------------------------------
@@ -104,7 +133,7 @@ DEV const bf16* b_chunk_addr(const ConvParams& p, long row, long kkg) {
 // Tile geometry is templated so small output-channel counts get a
 // narrow N tile instead of half-empty MFMA work: 4 waves arranged
 // (WR x WC) each owning a (BM_/WR x BN_/WC) sub-tile.
-template <bool DGRAD, int BM_, int BN_, int WR, int WC>
+template <int MODE, int BM_, int BN_, int WR, int WC>
 __global__ __launch_bounds__(THREADS, 2)
 void conv_igemm_kernel(ConvParams p) {
   constexpr int WM = BM_ / WR;        // wave tile M
@@ -164,8 +193,8 @@ void conv_igemm_kernel(ConvParams p) {
 #pragma unroll
     for (int l = 0; l < CA; ++l) {
       const long kkga = kk0 + a_cg[l] * 8;
-      const bf16* src = a_chunk_addr<DGRAD>(p, bm + a_row[l],
-                                            kkga < p.Kd ? kkga : 0);
+      const bf16* src = a_chunk_addr<MODE>(p, bm + a_row[l],
+                                           kkga < p.Kd ? kkga : 0);
       if (kkga >= p.Kd) src = p.zero;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
@@ -174,8 +203,8 @@ void conv_igemm_kernel(ConvParams p) {
     }
 #pragma unroll
     for (int l = 0; l < CB; ++l) {
-      const bf16* src = b_chunk_addr(p, bn + b_row[l],
-                                     kk0 + b_cg[l] * 8);
+      const bf16* src = b_chunk_addr<MODE>(p, bn + b_row[l],
+                                           kk0 + b_cg[l] * 8);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(lb +
@@ -240,11 +269,21 @@ void conv_igemm_kernel(ConvParams p) {
     for (int reg = 0; reg < 4; ++reg) {
       const long row = bm + wm + mf * 16 + fk * 4 + reg;
       if (row >= p.M) continue;
+      long out_row = row;
+      if (MODE == 2) {
+        // class row -> scattered dx pixel (n, 2*ii+a, 2*jj+b)
+        const int hw = p.nh * p.nw;
+        const int n = (int)(row / hw);
+        const int rem = (int)(row - (long)n * hw);
+        const int ii = rem / p.nw, jj = rem - (rem / p.nw) * p.nw;
+        out_row = ((long)n * p.H + 2 * ii + p.cls_a) * p.W
+                  + 2 * jj + p.cls_b;
+      }
 #pragma unroll
       for (int nf = 0; nf < NF; ++nf) {
         const long col = bn + wn + nf * 16 + fr;
         if (col < p.Nd)
-          p.out[row * p.Nd + col] = from_f32<bf16>(acc[mf][nf][reg]);
+          p.out[out_row * p.Nd + col] = from_f32<bf16>(acc[mf][nf][reg]);
       }
     }
   }
@@ -272,26 +311,47 @@ void launch_conv_igemm(const void* a, const void* b, void* out,
     p.Kd = (long)R * S * K;
   }
 
-#define LAUNCH(DG, BM_, BN_, WR, WC)                                        \
+#define LAUNCH(MODE, BM_, BN_, WR, WC)                                      \
   do {                                                                      \
     const long nbm = (p.M + (BM_) - 1) / (BM_);                             \
     const long nbn = (p.Nd + (BN_) - 1) / (BN_);                            \
     const size_t lds_bytes = 2 * ((BM_) + (BN_)) * BK * sizeof(bf16);       \
-    hipLaunchKernelGGL((conv_igemm_kernel<DG, BM_, BN_, WR, WC>),           \
+    hipLaunchKernelGGL((conv_igemm_kernel<MODE, BM_, BN_, WR, WC>),         \
                        dim3((unsigned)(nbm * nbn)), dim3(THREADS),          \
                        lds_bytes, stream, p);                               \
   } while (0)
+#define LAUNCH_TILED(MODE)                                                  \
+  do {                                                                      \
+    if (p.Nd <= 64 && p.M >= 256) LAUNCH(MODE, 256, 64, 4, 1);              \
+    else if (p.Nd <= 64) LAUNCH(MODE, 128, 64, 4, 1);                       \
+    else LAUNCH(MODE, 128, 128, 2, 2);                                      \
+  } while (0)
 
-  // narrow-N tiles when the output-channel dim can't fill 128 columns
   if (!dgrad) {
-    if (p.Nd <= 64 && p.M >= 256) LAUNCH(false, 256, 64, 4, 1);
-    else if (p.Nd <= 64) LAUNCH(false, 128, 64, 4, 1);
-    else LAUNCH(false, 128, 128, 2, 2);
+    LAUNCH_TILED(0);
+  } else if (stride == 2 && K % 8 == 0) {
+    // four parity classes, each a dense reduction over its valid taps
+    for (int a_ = 0; a_ < 2; ++a_) {
+      for (int b_ = 0; b_ < 2; ++b_) {
+        p.cls_a = a_;
+        p.cls_b = b_;
+        // r valid iff (ih + pad - r) even  ->  r ≡ (a_ + pad) (mod 2)
+        p.r0 = (a_ + pad) & 1;
+        p.s0 = (b_ + pad) & 1;
+        p.nr = (R - p.r0 + 1) / 2;
+        p.ns = (S - p.s0 + 1) / 2;
+        p.nh = (H - a_ + 1) / 2;
+        p.nw = (W - b_ + 1) / 2;
+        if (p.nr <= 0 || p.ns <= 0 || p.nh <= 0 || p.nw <= 0) continue;
+        p.M = (long)N * p.nh * p.nw;
+        p.Kd = (long)p.nr * p.ns * K;
+        LAUNCH_TILED(2);
+      }
+    }
   } else {
-    if (p.Nd <= 64 && p.M >= 256) LAUNCH(true, 256, 64, 4, 1);
-    else if (p.Nd <= 64) LAUNCH(true, 128, 64, 4, 1);
-    else LAUNCH(true, 128, 128, 2, 2);
+    LAUNCH_TILED(1);
   }
+#undef LAUNCH_TILED
 #undef LAUNCH
   HIP_CHECK_LAST();
 }

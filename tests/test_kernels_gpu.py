@@ -213,6 +213,7 @@ def test_depthwise_conv(dtype, tol, stride):
     (4, 64, 14, 14, 128, 1, 1, 0),      # 1x1
     (4, 64, 14, 14, 64, 3, 1, 1),       # 3x3 s1
     (4, 64, 15, 15, 128, 3, 2, 1),      # 3x3 s2, odd spatial
+    (4, 64, 14, 14, 128, 1, 2, 0),      # 1x1 s2 (degenerate dgrad classes)
     (2, 128, 7, 7, 120, 3, 1, 1),       # Nd not multiple of tile
     (2, 16, 9, 9, 24, 3, 2, 1),         # small C/K (mobilenet-ish)
 ])
