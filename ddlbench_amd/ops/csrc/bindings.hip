@@ -335,13 +335,12 @@ torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
   TORCH_CHECK(K % 8 == 0, "conv_igemm dgrad needs K % 8 == 0");
   // stride-2 parity classes with R==1 or S==1 have pixel classes no tap
   // reaches — those dx entries must be zeros, so zero-init in that case
-  auto dx = (stride == 2 && (R == 1 || S == 1))
-      ? torch::zeros({N, C, H, W},
-                     dy.options().memory_format(
-                         at::MemoryFormat::ChannelsLast))
-      : torch::empty({N, C, H, W},
-                     dy.options().memory_format(
-                         at::MemoryFormat::ChannelsLast));
+  // (zero_() after empty: torch::zeros does not honour the
+  // channels_last memory_format request)
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(
+                             at::MemoryFormat::ChannelsLast));
+  if (stride == 2 && (R == 1 || S == 1)) dx.zero_();
   auto zero = torch::zeros({16}, dy.options());
   launch_conv_igemm(dy.data_ptr(), w_perm.data_ptr(), dx.data_ptr(),
                     zero.data_ptr(), (int)N, (int)H, (int)W, (int)C, K,
