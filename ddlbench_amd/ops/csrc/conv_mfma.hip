@@ -60,54 +60,73 @@ struct ConvParams {
 
 // MODE: 0 = FPROP, 1 = DGRAD (any stride, gather with validity tests),
 //       2 = DGRAD stride-2 parity class (all taps valid by construction)
+//
+// The A operand's row -> pixel decomposition is loop-invariant (the
+// GEMM row never changes across K-steps), so it is computed once per
+// staging slot (a_row_coords) and only the (r, s, c/k) part runs per
+// step (a_step_addr).
+struct RowCoords { int n, y, x; bool valid; };
+
 template <int MODE>
-DEV const bf16* a_chunk_addr(const ConvParams& p, long m, long kkg) {
-  if (m >= p.M) return p.zero;
+DEV RowCoords a_row_coords(const ConvParams& p, long m) {
+  RowCoords rc;
+  rc.valid = m < p.M;
+  const long mm = rc.valid ? m : 0;
   if (MODE == 0) {
-    // m -> (n, oh, ow); kkg -> (r, s, c0)
     const int ohw = p.OH * p.OW;
-    const int n = (int)(m / ohw);
-    const int rem = (int)(m - (long)n * ohw);
-    const int oh = rem / p.OW, ow = rem - (rem / p.OW) * p.OW;
+    rc.n = (int)(mm / ohw);
+    const int rem = (int)(mm - (long)rc.n * ohw);
+    rc.y = rem / p.OW;
+    rc.x = rem - rc.y * p.OW;
+  } else if (MODE == 1) {
+    const int hw = p.H * p.W;
+    rc.n = (int)(mm / hw);
+    const int rem = (int)(mm - (long)rc.n * hw);
+    rc.y = rem / p.W;
+    rc.x = rem - rc.y * p.W;
+  } else {
+    const int hw = p.nh * p.nw;
+    rc.n = (int)(mm / hw);
+    const int rem = (int)(mm - (long)rc.n * hw);
+    const int ii = rem / p.nw, jj = rem - (rem / p.nw) * p.nw;
+    rc.y = 2 * ii + p.cls_a;
+    rc.x = 2 * jj + p.cls_b;
+  }
+  return rc;
+}
+
+template <int MODE>
+DEV const bf16* a_step_addr(const ConvParams& p, const RowCoords& rc,
+                            long kkg) {
+  if (!rc.valid) return p.zero;
+  if (MODE == 0) {
     const int c0 = (int)(kkg % p.Cin);
     const int rs = (int)(kkg / p.Cin);
     const int r = rs / p.S, s = rs - (rs / p.S) * p.S;
-    const int ih = oh * p.stride + r - p.pad;
-    const int iw = ow * p.stride + s - p.pad;
+    const int ih = rc.y * p.stride + r - p.pad;
+    const int iw = rc.x * p.stride + s - p.pad;
     if (ih < 0 || ih >= p.H || iw < 0 || iw >= p.W) return p.zero;
-    return p.a + (((long)n * p.H + ih) * p.W + iw) * p.Cin + c0;
+    return p.a + (((long)rc.n * p.H + ih) * p.W + iw) * p.Cin + c0;
   } else if (MODE == 1) {
-    // m -> (n, ih, iw); kkg -> (r, s, k0); gather from dy
-    const int hw = p.H * p.W;
-    const int n = (int)(m / hw);
-    const int rem = (int)(m - (long)n * hw);
-    const int ih = rem / p.W, iw = rem - (rem / p.W) * p.W;
     const int k0 = (int)(kkg % p.K);
     const int rs = (int)(kkg / p.K);
     const int r = rs / p.S, s = rs - (rs / p.S) * p.S;
-    const int tih = ih + p.pad - r;
-    const int tiw = iw + p.pad - s;
+    const int tih = rc.y + p.pad - r;
+    const int tiw = rc.x + p.pad - s;
     if (tih < 0 || tiw < 0 || (tih % p.stride) || (tiw % p.stride))
       return p.zero;
     const int oh = tih / p.stride, ow = tiw / p.stride;
     if (oh >= p.OH || ow >= p.OW) return p.zero;
-    return p.a + (((long)n * p.OH + oh) * p.OW + ow) * p.K + k0;
+    return p.a + (((long)rc.n * p.OH + oh) * p.OW + ow) * p.K + k0;
   } else {
-    // stride-2 class: m -> (n, ii, jj); ih = 2*ii+cls_a; only the class's
-    // valid (r, s) taps are enumerated, so no divisibility test
-    const int hw = p.nh * p.nw;
-    const int n = (int)(m / hw);
-    const int rem = (int)(m - (long)n * hw);
-    const int ii = rem / p.nw, jj = rem - (rem / p.nw) * p.nw;
-    const int ih = 2 * ii + p.cls_a, iw = 2 * jj + p.cls_b;
     const int k0 = (int)(kkg % p.K);
     const int rs = (int)(kkg / p.K);
     const int ri = rs / p.ns, si = rs - (rs / p.ns) * p.ns;
     const int r = p.r0 + 2 * ri, s = p.s0 + 2 * si;
-    const int oh = (ih + p.pad - r) >> 1;
-    const int ow = (iw + p.pad - s) >> 1;
+    const int oh = (rc.y + p.pad - r) >> 1;
+    const int ow = (rc.x + p.pad - s) >> 1;
     if (oh < 0 || ow < 0 || oh >= p.OH || ow >= p.OW) return p.zero;
-    return p.a + (((long)n * p.OH + oh) * p.OW + ow) * p.K + k0;
+    return p.a + (((long)rc.n * p.OH + oh) * p.OW + ow) * p.K + k0;
   }
 }
 
@@ -171,12 +190,14 @@ void conv_igemm_kernel(ConvParams p) {
 
   // ---- staging slots: slot ca -> (row = ca>>3, stored cg = ca&7),
   // holding source logical cg = (ca&7) ^ (row&7)
-  int a_row[CA], a_cg[CA], b_row[CB], b_cg[CB];
+  int a_cg[CA], b_row[CB], b_cg[CB];
+  RowCoords a_rc[CA];
 #pragma unroll
   for (int l = 0; l < CA; ++l) {
     const int ca = l * THREADS + tid;
-    a_row[l] = ca >> 3;
-    a_cg[l] = (ca & 7) ^ (a_row[l] & 7);
+    const int arow = ca >> 3;
+    a_cg[l] = (ca & 7) ^ (arow & 7);
+    a_rc[l] = a_row_coords<MODE>(p, bm + arow);
   }
 #pragma unroll
   for (int l = 0; l < CB; ++l) {
@@ -192,8 +213,8 @@ void conv_igemm_kernel(ConvParams p) {
 #pragma unroll
     for (int l = 0; l < CA; ++l) {
       const long kkga = kk0 + a_cg[l] * 8;
-      const bf16* src = a_chunk_addr<MODE>(p, bm + a_row[l],
-                                           kkga < p.Kd ? kkga : 0);
+      const bf16* src = a_step_addr<MODE>(p, a_rc[l],
+                                          kkga < p.Kd ? kkga : 0);
       if (kkga >= p.Kd) src = p.zero;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
