@@ -138,3 +138,139 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
     log.final(score, avg_sps, avg_secs)
     return {"bleu": score, "samples_per_sec": avg_sps,
             "sec_per_epoch": avg_secs, "valid_loss": val_loss}
+
+
+def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
+                      vocab=32320, hidden=1024, layers=4, lr=2.5e-4,
+                      dtype="float32", device="auto", seed=42,
+                      src_len_max=48, tgt_len=48, log_interval=0) -> dict:
+    """GNMT through the 1F1B pipeline — the reference's
+    translation/main_with_runtime.py flow on the
+    tuple-I/O StageRuntime. Straight pipeline, one stage per rank;
+    fixed (padded) sequence lengths give static edge shapes."""
+    import torch.distributed as dist
+
+    from ddlbench_amd.models.gnmt import (GNMT, LabelSmoothingLoss,
+                                          gnmt_edge_specs,
+                                          gnmt_pipeline_units)
+    from ddlbench_amd.parallel import init_distributed
+    from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
+    from ddlbench_amd.parallel.pipeline.balance import partition_minmax
+    from ddlbench_amd.parallel.pipeline.runtime import (StagePlan,
+                                                        StageRuntime)
+    from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
+
+    env = init_distributed()
+    world = env.world_size
+
+    class _C:
+        pass
+
+    cfg = _C()
+    cfg.device = device
+    dev = resolve_device(cfg, env.local_rank)
+    dt = torch.bfloat16 if dtype == "bfloat16" else torch.float32
+
+    torch.manual_seed(seed)
+    model = GNMT(vocab_size=vocab, hidden_size=hidden, num_layers=layers,
+                 dropout=0.0)
+    if dt != torch.float32:
+        model = model.to(dt)
+    units = gnmt_pipeline_units(model)
+    specs = gnmt_edge_specs(model, len(units), src_len_max, tgt_len,
+                            batch_size, dt)
+
+    # even split by unit count (decoder layers dominate roughly equally)
+    sizes = partition_minmax([1.0] * len(units), world)
+    bounds = [0]
+    for sz in sizes:
+        bounds.append(bounds[-1] + sz)
+    plan = StagePlan(replicas=[1] * world)
+    stage = env.rank
+    my_units = units[bounds[stage]:bounds[stage + 1]]
+
+    class _Tuple(torch.nn.Module):
+        def __init__(self, mods):
+            super().__init__()
+            self.mods = torch.nn.ModuleList(mods)
+
+        def forward(self, *xs):
+            for m in self.mods:
+                out = m(*xs)
+                xs = (out,) if torch.is_tensor(out) else out
+            return xs
+
+    stage_mod = _Tuple(my_units).to(dev)
+    transport = PipelineTransport(plan.edges(),
+                                  dist.get_backend() if world > 1
+                                  else "gloo")
+    in_specs = None if stage == 0 else specs[bounds[stage] - 1]
+    out_specs = specs[bounds[stage + 1] - 1]
+    loss_fn = LabelSmoothingLoss()
+    rt = StageRuntime(plan, env.rank, stage_mod, transport, in_specs,
+                      out_specs, dev, dt,
+                      loss_fn=lambda out, tgt: loss_fn(out, tgt))
+    opt = VersionedOptimizer(
+        FusedSGD(stage_mod.parameters(), lr=lr, momentum=0.9),
+        versioned=plan.num_warmup(stage) > 0)
+
+    def providers(epoch):
+        def tgt_full(mb):
+            g = torch.Generator().manual_seed(
+                seed * 131 + epoch * 7919 + mb)
+            return torch.randint(3, vocab, (tgt_len + 1, batch_size),
+                                 generator=g)
+
+        def input_provider(mb):
+            g = torch.Generator().manual_seed(
+                seed * 131 + epoch * 7919 + mb + 500_009)
+            src = torch.randint(3, vocab, (src_len_max, batch_size),
+                                generator=g)
+            src_len = torch.randint(src_len_max // 2, src_len_max + 1,
+                                    (batch_size,), generator=g)
+            return src, src_len, tgt_full(mb)[:-1]
+
+        def target_provider(mb):
+            return tgt_full(mb)[1:]
+
+        return input_provider, target_provider
+
+    log = BenchLogger(0 if rt.is_last else 1)
+    epoch_sps = []
+    last_loss = 0.0
+    for epoch in range(1, epochs + 1):
+        inp, tgt = providers(epoch)
+        mbs = rt.my_minibatches(n_minibatches)
+        warmup = min(plan.num_warmup(stage), len(mbs))
+        losses = []
+        if dist.is_initialized():
+            dist.barrier()
+        if dev.type == "cuda":
+            torch.cuda.synchronize(dev)
+        t0 = time.perf_counter()
+        for k in range(warmup):
+            loss, _ = rt.run_forward(mbs[k], inp, tgt, training=True)
+            if loss is not None:
+                losses.append(loss.detach())
+        for k in range(len(mbs)):
+            if warmup + k < len(mbs):
+                loss, _ = rt.run_forward(mbs[warmup + k], inp, tgt,
+                                         training=True)
+                if loss is not None:
+                    losses.append(loss.detach())
+            opt.zero_grad(set_to_none=False)
+            rt.run_backward()
+            opt.step()
+        if dev.type == "cuda":
+            torch.cuda.synchronize(dev)
+        if dist.is_initialized():
+            dist.barrier()
+        secs = time.perf_counter() - t0
+        sps = n_minibatches * batch_size / secs
+        epoch_sps.append(sps)
+        last_loss = (torch.stack(losses).mean().item() if losses else 0.0)
+        log.epoch(epoch, epochs, last_loss, sps, last_loss, 0.0)
+    avg = sum(epoch_sps) / max(len(epoch_sps), 1)
+    log.final(0.0, avg, 0.0)
+    return {"samples_per_sec": avg, "train_loss": last_loss,
+            "stage": stage, "num_units": len(units)}

@@ -228,6 +228,129 @@ class GNMT(nn.Module):
         return logits  # (Ttgt, B, V)
 
 
+# ---------------------------------------------------- pipeline stages
+class _EncEmbed(nn.Module):
+    """(src, src_len, tgt_in) -> (x, src_len, tgt_in)"""
+
+    def __init__(self, enc: ResidualRecurrentEncoder):
+        super().__init__()
+        self.embedder = enc.embedder
+        self.bidir = enc.bidir
+        self.dropout = enc.dropout
+
+    def forward(self, src, src_len, tgt_in):
+        x = self.dropout(self.embedder(src.long()))
+        x = self.bidir(x, src_len.long())
+        return x, src_len, tgt_in
+
+
+class _EncLayer1(nn.Module):
+    def __init__(self, enc):
+        super().__init__()
+        self.layer1 = enc.layer1
+        self.dropout = enc.dropout
+
+    def forward(self, x, src_len, tgt_in):
+        y, _ = self.layer1(self.dropout(x))
+        return y, src_len, tgt_in
+
+
+class _EncResidual(nn.Module):
+    def __init__(self, rnn, dropout):
+        super().__init__()
+        self.rnn = rnn
+        self.dropout = dropout
+
+    def forward(self, x, src_len, tgt_in):
+        y, _ = self.rnn(self.dropout(x))
+        return x + y, src_len, tgt_in
+
+
+class _DecAttn(nn.Module):
+    """(context, src_len, tgt_in) -> (y, ctx, tgt_in)"""
+
+    def __init__(self, dec: ResidualRecurrentDecoder):
+        super().__init__()
+        self.embedder = dec.embedder
+        self.att_rnn = dec.att_rnn
+
+    def forward(self, context, src_len, tgt_in):
+        mask = varlen_mask(src_len.long(), context.size(0)).transpose(0, 1)
+        x = self.embedder(tgt_in.long())
+        y, ctx, _, _ = self.att_rnn(x, context, mask)
+        return y, ctx
+
+
+class _DecLayer1(nn.Module):
+    def __init__(self, dec):
+        super().__init__()
+        self.layer1 = dec.layer1
+        self.dropout = dec.dropout
+
+    def forward(self, y, ctx):
+        z, _ = self.layer1(self.dropout(torch.cat([y, ctx], dim=2)))
+        return z, ctx
+
+
+class _DecResidual(nn.Module):
+    def __init__(self, rnn, dropout):
+        super().__init__()
+        self.rnn = rnn
+        self.dropout = dropout
+
+    def forward(self, y, ctx):
+        z, _ = self.rnn(torch.cat([self.dropout(y), ctx], dim=2))
+        return y + z, ctx
+
+
+class _DecHead(nn.Module):
+    def __init__(self, dec):
+        super().__init__()
+        self.classifier = dec.classifier
+
+    def forward(self, y, ctx):
+        return self.classifier(y)
+
+
+def gnmt_pipeline_units(model: GNMT):
+    """Flatten GNMT into tuple-I/O pipeline units (the counterpart of
+    the reference's generated gnmt stage modules,
+    pipedream-fork/runtime/translation/models/gnmt/gpus=4/). LSTM hidden
+    state never crosses a unit boundary (each minibatch starts fresh),
+    so any contiguous grouping of units is a valid stage."""
+    enc, dec = model.encoder, model.decoder
+    units = [_EncEmbed(enc), _EncLayer1(enc)]
+    units += [_EncResidual(r, enc.dropout) for r in enc.layers]
+    units += [_DecAttn(dec), _DecLayer1(dec)]
+    units += [_DecResidual(r, dec.dropout) for r in dec.layers]
+    units += [_DecHead(dec)]
+    return units
+
+
+def gnmt_edge_specs(model: GNMT, n_units: int, Tsrc: int, Ttgt: int,
+                    B: int, dtype):
+    """Static TensorSpec lists for every unit boundary, given fixed
+    (padded) sequence lengths — pipeline mode trades the bucketing
+    sampler's ragged batches for static shapes."""
+    from ddlbench_amd.parallel.pipeline.runtime import TensorSpec
+    H = model.encoder.layer1.hidden_size
+    n_enc = 2 + len(model.encoder.layers)   # units producing (x,len,tgt)
+    specs = []
+    for i in range(n_units):
+        if i < n_enc:  # after unit i: (x, src_len, tgt_in)
+            specs.append([
+                TensorSpec((Tsrc, B, 2 * H if i == 0 else H), dtype, True),
+                TensorSpec((B,), torch.long, False),
+                TensorSpec((Ttgt, B), torch.long, False)])
+        elif i < n_units - 1:  # decoder body: (y, ctx)
+            specs.append([TensorSpec((Ttgt, B, H), dtype, True),
+                          TensorSpec((Ttgt, B, H), dtype, True)])
+        else:  # logits
+            specs.append([TensorSpec((Ttgt, B, model.vocab_size), dtype,
+                                     True)])
+    return specs
+
+
 class LabelSmoothingLoss(nn.Module):
     """Per-token label-smoothed CE ignoring PAD
     (reference train/smoothing.py:7-18)."""

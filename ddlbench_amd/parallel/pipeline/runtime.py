@@ -21,7 +21,7 @@ from __future__ import annotations
 import math
 from collections import deque
 from dataclasses import dataclass
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Optional, Sequence, Tuple, Union
 
 import torch
 import torch.distributed as dist
@@ -83,6 +83,28 @@ class StagePlan:
         return max(after // self.replicas[stage], 0)
 
 
+@dataclass
+class TensorSpec:
+    """Static shape/dtype of one tensor on a pipeline edge. ``grad``
+    marks float tensors whose gradient travels back on the bwd channel;
+    pass-through tensors (lengths, token ids, masks) set grad=False and
+    ride the pipeline forward only (the reference routes e.g.
+    target_length the same way, runtime.py:540-543)."""
+    shape: Tuple[int, ...]
+    dtype: "torch.dtype" = None
+    grad: bool = True
+
+
+def _as_specs(x, default_dtype) -> Optional[List[TensorSpec]]:
+    """Accept torch.Size (single-tensor API) or a list of TensorSpec."""
+    if x is None:
+        return None
+    if isinstance(x, (list, tuple)) and x and isinstance(x[0], TensorSpec):
+        return [TensorSpec(tuple(t.shape), t.dtype or default_dtype, t.grad)
+                for t in x]
+    return [TensorSpec(tuple(x), default_dtype, True)]
+
+
 class RuntimeStats:
     """fwd/bwd counters: compute time, send/recv bytes (the reference's
     runtime_utilities.RuntimeStats)."""
@@ -106,8 +128,8 @@ class StageRuntime:
     def __init__(self, plan: StagePlan, rank: int,
                  module: torch.nn.Module,
                  transport: PipelineTransport,
-                 in_shape: Optional[torch.Size],
-                 out_shape: torch.Size,
+                 in_shape,   # torch.Size | List[TensorSpec] | None
+                 out_shape,  # torch.Size | List[TensorSpec]
                  device: torch.device,
                  dtype: torch.dtype,
                  loss_fn: Optional[Callable] = None,
@@ -117,8 +139,8 @@ class StageRuntime:
         self.stage, self.replica = plan.stage_of_rank(rank)
         self.module = module
         self.transport = transport
-        self.in_shape = in_shape
-        self.out_shape = out_shape
+        self.in_specs = _as_specs(in_shape, dtype)
+        self.out_specs = _as_specs(out_shape, dtype)
         self.device = device
         self.dtype = dtype
         self.loss_fn = loss_fn
@@ -146,68 +168,91 @@ class StageRuntime:
     def run_forward(self, mb: int, input_provider=None,
                     target_provider=None, training: bool = True):
         if self.is_first:
-            x = input_provider(mb).to(self.device, dtype=self.dtype,
-                                      non_blocking=True)
+            xs = input_provider(mb)
+            if torch.is_tensor(xs):
+                xs = (xs,)
+            xs = tuple(
+                t.to(self.device,
+                     dtype=self.dtype if t.is_floating_point() else None,
+                     non_blocking=True) for t in xs)
         else:
-            buf = torch.empty(self.in_shape, device=self.device,
-                              dtype=self.dtype)
             ch = self.transport.channel(self._prev_rank(mb), self.rank,
                                         "fwd")
-            ch.irecv(buf).wait()
-            self.stats.recv_bytes += buf.numel() * buf.element_size()
-            x = buf
+            bufs = []
+            for spec in self.in_specs:
+                buf = torch.empty(spec.shape, device=self.device,
+                                  dtype=spec.dtype)
+                ch.irecv(buf).wait()
+                self.stats.recv_bytes += buf.numel() * buf.element_size()
+                bufs.append(buf)
+            xs = tuple(bufs)
         if training and not self.is_first:
-            x.requires_grad_(True)
+            for t, spec in zip(xs, self.in_specs):
+                if spec.grad:
+                    t.requires_grad_(True)
         with torch.enable_grad() if training else torch.no_grad():
-            out = self.module(x)
+            out = self.module(*xs)
+        outs = (out,) if torch.is_tensor(out) else tuple(out)
         send_work = None
         loss = None
         extras = {}
         if self.is_last:
             if self.loss_fn is not None and target_provider is not None:
                 y = target_provider(mb).to(self.device, non_blocking=True)
-                loss = self.loss_fn(out, y)
+                loss = self.loss_fn(outs[0] if len(outs) == 1 else outs, y)
                 extras["target"] = y
-                extras["output"] = out.detach()
+                extras["output"] = outs[0].detach()
         else:
             ch = self.transport.channel(self.rank, self._next_rank(mb),
                                         "fwd")
-            send_work = ch.isend(out.detach())
-            self.stats.send_bytes += out.numel() * out.element_size()
-        self.inflight.append((x, loss if self.is_last else out, send_work,
-                              mb, extras))
+            send_work = [ch.isend(t.detach()) for t in outs]
+            for t in outs:
+                self.stats.send_bytes += t.numel() * t.element_size()
+        self.inflight.append((xs, loss if self.is_last else outs,
+                              send_work, mb, extras))
         self.stats.fwd_count += 1
         return loss, extras
 
     def run_backward(self):
-        x, out, send_work, mb, _ = self.inflight.popleft()
+        xs, out, send_work, mb, _ = self.inflight.popleft()
         if send_work is not None:
-            send_work.wait()  # activation must be delivered before reuse
+            for w in send_work:
+                w.wait()  # activations must be delivered before reuse
         if self.is_last:
             out.backward()  # out is the loss
         else:
-            gbuf = torch.empty(self.out_shape, device=self.device,
-                               dtype=self.dtype)
             ch = self.transport.channel(self.rank, self._next_rank(mb),
                                         "bwd")
-            ch.irecv(gbuf).wait()
-            out.backward(gbuf)
+            grads, grad_outs = [], []
+            for t, spec in zip(out, self.out_specs):
+                if not spec.grad:
+                    continue
+                gbuf = torch.empty(spec.shape, device=self.device,
+                                   dtype=spec.dtype)
+                ch.irecv(gbuf).wait()
+                grads.append(gbuf)
+                grad_outs.append(t)
+            torch.autograd.backward(grad_outs, grads)
         if not self.is_first:
             ch = self.transport.channel(self._prev_rank(mb), self.rank,
                                         "bwd")
-            # send input grad upstream; wait so the grad buffer lifetime
-            # is safe (upstream is blocked on it anyway — the pipeline's
-            # critical path is unaffected)
-            ch.isend(x.grad).wait()
-            self.stats.send_bytes += x.grad.numel() * x.grad.element_size()
+            for t, spec in zip(xs, self.in_specs):
+                if not spec.grad:
+                    continue
+                g = t.grad if t.grad is not None else torch.zeros_like(t)
+                # wait so the grad buffer lifetime is safe (upstream is
+                # blocked on it anyway)
+                ch.isend(g).wait()
+                self.stats.send_bytes += g.numel() * g.element_size()
         self.stats.bwd_count += 1
         return mb
 
     def pop_eval(self):
-        """Retire the oldest eval-mode forward (wait for its send)."""
+        """Retire the oldest eval-mode forward (wait for its sends)."""
         x, out, send_work, mb, extras = self.inflight.popleft()
         if send_work is not None:
-            send_work.wait()
+            for w in send_work:
+                w.wait()
         return mb, extras
 
     # ---- eval ack clocking (reference run_ack, runtime.py:630-654) ----

@@ -195,3 +195,27 @@ def _worker_hybrid(rank, world, port):
 @pytest.mark.timeout(300)
 def test_hybrid_replicated_stage(free_port):
     mp.spawn(_worker_hybrid, args=(3, free_port), nprocs=3, join=True)
+
+
+# --------------------------------------- GNMT through the 1F1B pipeline
+def _worker_gnmt_pipe(rank, world, port):
+    _env(rank, world, port)
+    from ddlbench_amd.gnmt_runner import run_gnmt_pipeline
+    res = run_gnmt_pipeline(epochs=1, batch_size=4, n_minibatches=6,
+                            vocab=64, hidden=16, layers=4,
+                            device="cpu", src_len_max=10, tgt_len=9)
+    assert res["samples_per_sec"] > 0
+    if rank == world - 1:
+        assert res["train_loss"] > 0
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_gnmt_1f1b_pipeline_two_stages(free_port):
+    mp.spawn(_worker_gnmt_pipe, args=(2, free_port), nprocs=2, join=True)
+
+
+@pytest.mark.timeout(600)
+def test_gnmt_1f1b_pipeline_three_stages(free_port):
+    mp.spawn(_worker_gnmt_pipe, args=(3, free_port), nprocs=3, join=True)
