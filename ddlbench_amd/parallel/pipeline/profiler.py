@@ -26,6 +26,64 @@ def _sync(device: torch.device) -> None:
         torch.cuda.synchronize(device)
 
 
+def trace_module_graph(model: nn.Module, *sample_args) -> Graph:
+    """Build the layer DAG of an arbitrary model from one traced forward.
+
+    The reference wraps tensors and monkey-patches forwards to trace the
+    DAG (pipedream-fork/profiler/torchmodules/torchgraph/
+    graph_creator.py:213-288). Here plain forward hooks on leaf modules
+    suffice: a node per leaf-module call, edges by matching input tensor
+    identities to the node that produced them. Residual forks/joins are
+    visible at module level because the joins are fused module inputs
+    (BNAct(x, res)) rather than free-floating functional adds.
+
+    Nodes carry desc + parameter_size + activation_size; timing comes
+    from profile_sequential (chains) or can be merged in later."""
+    records = []  # (name, module, input_ids, output_ids, out_bytes)
+    hooks = []
+    keepalive = []  # hold refs so freed tensors can't recycle an id()
+
+    def hook(module, inputs, kwargs, output, name=""):
+        all_in = list(inputs) + list((kwargs or {}).values())
+        in_ids = [id(t) for t in all_in if torch.is_tensor(t)]
+        outs = output if isinstance(output, (tuple, list)) else (output,)
+        out_ids = [id(t) for t in outs if torch.is_tensor(t)]
+        nbytes = sum(t.numel() * t.element_size()
+                     for t in outs if torch.is_tensor(t))
+        keepalive.extend(t for t in outs if torch.is_tensor(t))
+        records.append((name, module, in_ids, out_ids, nbytes))
+
+    for name, m in model.named_modules():
+        if len(list(m.children())) == 0 and name:
+            hooks.append(m.register_forward_hook(
+                lambda mod, i, kw, o, name=name: hook(mod, i, kw, o, name),
+                with_kwargs=True))
+    was_training = model.training
+    model.eval()
+    try:
+        with torch.no_grad():
+            model(*sample_args)
+    finally:
+        for h in hooks:
+            h.remove()
+        model.train(was_training)
+
+    g = Graph()
+    producer = {}  # tensor id -> node id
+    for i, (name, m, in_ids, out_ids, nbytes) in enumerate(records):
+        pbytes = sum(p.numel() * p.element_size()
+                     for p in m.parameters(recurse=False))
+        g.add_node(Node(i, desc=f"{name}:{type(m).__name__}",
+                        activation_size=float(nbytes),
+                        parameter_size=float(pbytes)))
+        for tid in in_ids:
+            if tid in producer:
+                g.add_edge(producer[tid], i)
+        for tid in out_ids:
+            producer[tid] = i
+    return g
+
+
 def profile_sequential(seq: nn.Sequential, sample: torch.Tensor,
                        device: Optional[torch.device] = None,
                        iters: int = 8, warmup: int = 2) -> Graph:

@@ -96,3 +96,29 @@ def test_profiler_emits_chain():
     assert nodes[0].parameter_size == (8 * 16 + 16) * 4
     assert all(n.fwd_time >= 0 for n in nodes)
     assert nodes[-1].activation_size == 4 * 4 * 4
+
+
+def test_trace_module_graph_captures_residual_dag():
+    """Hook tracer recovers the skip-connection DAG of a resnet block."""
+    from ddlbench_amd.models import build_model
+    from ddlbench_amd.parallel.pipeline.profiler import trace_module_graph
+    m = build_model("cifar10", "resnet18")
+    g = trace_module_graph(m, torch.randn(1, 3, 32, 32))
+    assert len(g.nodes) > 30
+    # at least one join: a node with 2 in-edges (BNAct taking x + res)
+    joins = [i for i in g.nodes if len(g.in_edges[i]) >= 2]
+    assert joins, "residual joins not captured"
+    # at least one fork: a node feeding 2 consumers
+    forks = [i for i in g.nodes if len(g.edges[i]) >= 2]
+    assert forks, "residual forks not captured"
+    assert not g.is_chain()
+    g.topological_sort()  # acyclic
+
+
+def test_trace_module_graph_chain_for_mlp():
+    from ddlbench_amd.parallel.pipeline.profiler import trace_module_graph
+    m = torch.nn.Sequential(torch.nn.Linear(4, 8), torch.nn.ReLU(),
+                            torch.nn.Linear(8, 2))
+    g = trace_module_graph(m, torch.randn(2, 4))
+    assert g.is_chain()
+    assert len(g.nodes) == 3
