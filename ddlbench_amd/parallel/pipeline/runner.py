@@ -123,8 +123,12 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
                       out_shape, device, dtype, loss_fn=loss_fn,
                       dp_wrapper=dp)
 
+    # per-stage LR scaling for replicated stages (reference
+    # runtime.py:695-701: DP-style x-replicas within a stage)
+    stage_lr = cfg.lr * plan.replicas[stage]
     opt = VersionedOptimizer(
-        FusedSGD(stage_mod.parameters(), lr=cfg.lr, momentum=cfg.momentum,
+        FusedSGD(stage_mod.parameters(), lr=stage_lr,
+                 momentum=cfg.momentum,
                  weight_decay=cfg.weight_decay,
                  backend=cfg.kernel_backend),
         versioned=(not cfg.no_input_pipelining
