@@ -11,11 +11,17 @@ Eligibility: bf16, channels_last, groups=1, dilation=1, C%8==0, K%8==0.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from ddlbench_amd import ops as _ops
+
+# weight-grad backend: "mfma" runs the native kernel (conv_wgrad.hip),
+# anything else the library conv backward
+_WGRAD = os.environ.get("DDLB_WGRAD", "library")
 
 _CL = torch.channels_last
 
@@ -45,11 +51,19 @@ class _ConvMFMA(torch.autograd.Function):
                                       x.size(2), x.size(3), ctx.stride,
                                       ctx.pad)
         if ctx.needs_input_grad[1]:
-            # library wgrad (native MFMA wgrad kernel: TODO)
-            dw = torch.ops.aten.convolution_backward(
-                dy, x, weight, None, [ctx.stride, ctx.stride],
-                [ctx.pad, ctx.pad], [1, 1], False, [0, 0], 1,
-                [False, True, False])[1]
+            if _WGRAD == "mfma":
+                R = weight.shape[2]
+                dw32 = ext.conv_igemm_wgrad(x, dy, R, weight.shape[3],
+                                            ctx.stride, ctx.pad)
+                # (K, R*S*C) memory -> logical (K,C,R,S) channels_last
+                dw = dw32.view(weight.shape[0], R, weight.shape[3],
+                               weight.shape[1]) \
+                    .permute(0, 3, 1, 2).to(weight.dtype)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    dy, x, weight, None, [ctx.stride, ctx.stride],
+                    [ctx.pad, ctx.pad], [1, 1], False, [0, 0], 1,
+                    [False, True, False])[1]
         return dx, dw, None, None
 
 

@@ -53,6 +53,9 @@ void launch_ce_bwd(const T*, const int64_t*, const float*, const float*, T*,
 void launch_conv_igemm(const void*, const void*, void*, const void*, int,
                        int, int, int, int, int, int, int, int, int, int,
                        int, hipStream_t);
+void launch_conv_wgrad(const void*, const void*, float*, const void*, int,
+                       int, int, int, int, int, int, int, int, int, int,
+                       hipStream_t);
 
 template <typename T>
 void launch_revert_varlen(const T*, T*, const int64_t*, int64_t, int64_t,
@@ -349,6 +352,31 @@ torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
   return dx;
 }
 
+// x: (N,C,H,W) channels_last; dy: (N,K,OH,OW) channels_last.
+// Returns fp32 dw in (K, R*S*C) memory = logical (K,C,R,S) channels_last
+// after the python-side permute.
+torch::Tensor conv_igemm_wgrad(torch::Tensor x, torch::Tensor dy,
+                               int64_t R, int64_t S, int64_t stride,
+                               int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "conv_wgrad: bf16 HIP tensors only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels_last");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "dy must be channels_last");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  TORCH_CHECK(C % 8 == 0 && K % 8 == 0,
+              "conv_wgrad needs C %% 8 == 0 and K %% 8 == 0");
+  auto dw = torch::zeros({K, R * S * C}, x.options().dtype(at::kFloat));
+  auto zero = torch::zeros({16}, x.options());
+  launch_conv_wgrad(x.data_ptr(), dy.data_ptr(),
+                    dw.data_ptr<float>(), zero.data_ptr(), N, H, W, C, K,
+                    OH, OW, (int)R, (int)S, (int)stride, (int)pad,
+                    cur_stream());
+  return dw;
+}
+
 // ---- depthwise 3x3 ----------------------------------------------------
 torch::Tensor dw3x3_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                         bool nhwc) {
@@ -445,6 +473,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC bf16 MFMA implicit-GEMM conv forward");
   m.def("conv_igemm_dgrad", &conv_igemm_dgrad,
         "NHWC bf16 MFMA implicit-GEMM conv data-grad");
+  m.def("conv_igemm_wgrad", &conv_igemm_wgrad,
+        "NHWC bf16 MFMA implicit-GEMM conv weight-grad (fp32 out)");
   m.def("ce_fwd", &ce_fwd, "cross-entropy forward");
   m.def("ce_bwd", &ce_bwd, "cross-entropy backward");
   m.def("revert_varlen", &revert_varlen,

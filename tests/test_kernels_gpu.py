@@ -245,6 +245,36 @@ def test_conv_mfma_fwd_dgrad(shape):
                                rtol=5e-2, atol=1.0)
 
 
+@pytest.mark.parametrize("shape", [
+    (4, 64, 14, 14, 64, 3, 1, 1),
+    (4, 64, 15, 15, 128, 3, 2, 1),
+    (2, 32, 9, 9, 24, 1, 1, 0),
+])
+def test_conv_wgrad_native(shape):
+    """Native MFMA weight-grad kernel vs the fp32 torch reference."""
+    from ddlbench_amd.ops import require_extension
+    ext = require_extension()
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(0)
+    dev = _dev()
+    x = torch.randn(N, C, H, W, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    OH = (H + 2 * pad - R) // stride + 1
+    dy = torch.randn(N, K, OH, OH, device=dev, dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    dw32 = ext.conv_igemm_wgrad(x, dy, R, R, stride, pad)
+    dw = dw32.view(K, R, R, C).permute(0, 3, 1, 2)
+
+    x2 = x.detach().float().contiguous().requires_grad_(True)
+    w2 = torch.zeros(K, C, R, R, device=dev, requires_grad=True)
+    y2 = torch.nn.functional.conv2d(x2, w2, None, stride, pad)
+    y2.backward(dy.float().contiguous())
+    # reduction over N*OH*OW in bf16 products: tolerance grows with P
+    p = N * OH * OH
+    tol = 0.02 * (p ** 0.5)
+    torch.testing.assert_close(dw.float(), w2.grad, rtol=5e-2, atol=tol)
+
+
 def test_conv_mfma_resnet_block_trains():
     """convert_convs on a resnet block; one train step, finite loss."""
     from ddlbench_amd.models import build_model
