@@ -88,6 +88,17 @@ def main():
             [0, 0], 1, [True, False, False]), args.iters, args.warmup)
         res["miopen_dgrad_ms"] = round(t * 1e3, 3)
         res["miopen_dgrad_tf"] = round(flops / t / 1e12, 1)
+        # wgrad
+        t = bench_op(lambda: ext.conv_igemm_wgrad(x, dy, R, R, stride,
+                                                  pad),
+                     args.iters, args.warmup)
+        res["mfma_wgrad_ms"] = round(t * 1e3, 3)
+        res["mfma_wgrad_tf"] = round(flops / t / 1e12, 1)
+        t = bench_op(lambda: torch.ops.aten.convolution_backward(
+            dy, x, w, None, [stride, stride], [pad, pad], [1, 1], False,
+            [0, 0], 1, [False, True, False]), args.iters, args.warmup)
+        res["miopen_wgrad_ms"] = round(t * 1e3, 3)
+        res["miopen_wgrad_tf"] = round(flops / t / 1e12, 1)
         print(json.dumps(res), flush=True)
 
 
