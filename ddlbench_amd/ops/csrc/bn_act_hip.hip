@@ -716,7 +716,7 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
                          i64max(2048 / cblocks, 1));
       hipLaunchKernelGGL((bn_stats_nhwc_vec_kernel<T>), dim3(cblocks, S),
                          dim3(block), 0, stream, x, sums, rows, C, CG8);
-    } else if (sizeof(T) == 4 && C % 4 == 0) {
+    } else if (sizeof(T) == 4 && C % 4 == 0 && g_bn_variant == 2) {
       const int CG4 = (int)i64min(C / 4, 64);
       const int64_t cblocks = (C / 4 + CG4 - 1) / CG4;
       int64_t S = i64min(i64max(rows / 512, 1),

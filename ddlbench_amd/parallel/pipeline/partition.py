@@ -88,15 +88,20 @@ def _dp_allreduce_time(r: int, param_bytes: float, bw: float) -> float:
 
 def partition_chain(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
                     memory_bytes: float = MI355X_MEM,
-                    straight: bool = False) -> PartitionResult:
+                    straight: bool = False,
+                    inference: bool = False) -> PartitionResult:
     """Optimal contiguous partition with per-stage replication.
 
     straight=True disables replication (pure pipeline, one GPU per
-    stage — the reference's --straight_pipeline)."""
+    stage — the reference's --straight_pipeline). inference=True
+    partitions a forward-only pipeline (the reference's
+    inference_optimizer_graph.py): stage time is forward time only,
+    activations cross each cut once, and there is no weight-stash
+    memory term."""
     nodes = graph.topological_sort()
     n = len(nodes)
     ids = [nd.node_id for nd in nodes]
-    t = [nd.compute_time for nd in nodes]
+    t = [nd.fwd_time if inference else nd.compute_time for nd in nodes]
     act = [nd.activation_size for nd in nodes]
     par = [nd.parameter_size for nd in nodes]
     # prefix sums
@@ -124,11 +129,14 @@ def partition_chain(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
                         continue
                     T = pt[i] - pt[j]
                     P = pp[i] - pp[j]
-                    stage_time = T / r + _dp_allreduce_time(r, P, bw)
-                    comm_in = (2.0 * act[j - 1] / (bw * r)) if j > 0 else 0.0
+                    stage_time = T / r + (0.0 if inference else
+                                          _dp_allreduce_time(r, P, bw))
+                    act_xfers = 1.0 if inference else 2.0
+                    comm_in = (act_xfers * act[j - 1] / (bw * r)) \
+                        if j > 0 else 0.0
                     # memory: stash depth ~ remaining pipeline depth; use
                     # the conservative machine-count bound like the ref
-                    stash = max(M - m + 1, 1)
+                    stash = 0 if inference else max(M - m + 1, 1)
                     act_bytes = pa[i] - pa[j]
                     if (stash + 1) * (P + act_bytes / max(r, 1)) > memory_bytes:
                         continue

@@ -122,3 +122,26 @@ def test_trace_module_graph_chain_for_mlp():
     g = trace_module_graph(m, torch.randn(2, 4))
     assert g.is_chain()
     assert len(g.nodes) == 3
+
+
+def test_inference_partition_uses_forward_time():
+    # heavy backward on layer 2 matters for training, not inference
+    g = Graph.chain([
+        Node(0, fwd_time=1.0, bwd_time=0.0, activation_size=1,
+             parameter_size=1),
+        Node(1, fwd_time=1.0, bwd_time=0.0, activation_size=1,
+             parameter_size=1),
+        Node(2, fwd_time=1.0, bwd_time=8.0, activation_size=1,
+             parameter_size=1),
+        Node(3, fwd_time=1.0, bwd_time=0.0, activation_size=1,
+             parameter_size=1),
+    ])
+    train = partition_chain(g, 2, straight=True)
+    assert any(s.layers == [2] for s in train.stages) or \
+        train.module_to_stage_map[2] != train.module_to_stage_map[1]
+    g2 = Graph.chain([
+        Node(i, fwd_time=ft, bwd_time=bw, activation_size=1,
+             parameter_size=1)
+        for i, (ft, bw) in enumerate([(1, 0), (1, 0), (1, 8), (1, 0)])])
+    infer = partition_chain(g2, 2, straight=True, inference=True)
+    assert infer.module_to_stage_map == [0, 0, 1, 1]
