@@ -2,7 +2,6 @@
 
 from __future__ import annotations
 
-import sys
 
 from ddlbench_amd.config import config_from_args, make_parser
 from ddlbench_amd.strategies import run

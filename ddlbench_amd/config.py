@@ -19,7 +19,6 @@ from __future__ import annotations
 import argparse
 import dataclasses
 import os
-from typing import Optional
 
 # Dataset shape table — mirrors the reference's synthetic generator
 # (/root/reference/benchmark/generate_synthetic_data.py:76-107).

@@ -13,7 +13,6 @@ import time
 from typing import Callable, Optional
 
 import torch
-import torch.nn.functional as F
 
 from ddlbench_amd.config import BenchConfig
 from ddlbench_amd.ops import functional as NF

@@ -12,7 +12,7 @@ import time
 
 import torch
 
-from ddlbench_amd.engine import compute_dtype, resolve_device
+from ddlbench_amd.engine import resolve_device
 from ddlbench_amd.models.gnmt import GNMT, LabelSmoothingLoss
 from ddlbench_amd.data.translation import (BucketingSampler,
                                            SyntheticTranslationDataset,

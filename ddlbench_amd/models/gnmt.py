@@ -12,7 +12,7 @@ HIP kernel (ops/csrc/seq_utils.hip) with a torch fallback on CPU."""
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn as nn

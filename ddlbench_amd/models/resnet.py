@@ -13,7 +13,6 @@ of three."""
 
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

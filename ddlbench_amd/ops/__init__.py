@@ -13,9 +13,6 @@ for gfx950 only (``python setup.py build_ext --inplace`` or
 
 from __future__ import annotations
 
-import importlib
-import os
-import sys
 
 _ext = None
 _ext_err: Exception | None = None

@@ -21,7 +21,7 @@ MI355X-native rebuild:
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 import torch.distributed as dist

@@ -30,8 +30,8 @@ convert_graph_to_model.py:559-586)."""
 from __future__ import annotations
 
 import json
-from dataclasses import dataclass, field
-from typing import List, Optional
+from dataclasses import dataclass
+from typing import List
 
 from ddlbench_amd.parallel.pipeline.graph import Graph
 

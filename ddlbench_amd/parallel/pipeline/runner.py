@@ -12,11 +12,8 @@ forward-only validation with ack clocking."""
 
 from __future__ import annotations
 
-import json
-import math
 import os
 import time
-from typing import List
 
 import torch
 import torch.distributed as dist
@@ -29,10 +26,8 @@ from ddlbench_amd.ops import functional as NF
 from ddlbench_amd.ops.modules import set_default_backend
 from ddlbench_amd.ops.sgd import FusedSGD
 from ddlbench_amd.parallel import BucketedDataParallel, init_distributed
-from ddlbench_amd.parallel.dist_utils import distributed_env
 from ddlbench_amd.parallel.pipeline.comm import (PipelineTransport,
                                                  dry_run_shapes)
-from ddlbench_amd.parallel.pipeline.graph import Graph
 from ddlbench_amd.parallel.pipeline.partition import (PartitionResult,
                                                       partition_chain)
 from ddlbench_amd.parallel.pipeline.profiler import profile_sequential

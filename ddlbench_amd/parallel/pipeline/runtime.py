@@ -21,10 +21,9 @@ from __future__ import annotations
 import math
 from collections import deque
 from dataclasses import dataclass
-from typing import Callable, List, Optional, Sequence, Tuple, Union
+from typing import Callable, List, Optional, Tuple
 
 import torch
-import torch.distributed as dist
 
 from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
 

@@ -14,7 +14,7 @@ from ddlbench_amd.config import BenchConfig
 from ddlbench_amd.data import make_loaders
 from ddlbench_amd.engine import (Trainer, compute_dtype, make_optimizer,
                                  resolve_device)
-from ddlbench_amd.models import build_model, build_sequential
+from ddlbench_amd.models import build_model
 from ddlbench_amd.utils import BenchLogger
 
 

@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import collections
 import math
-from typing import List, Sequence
+from typing import Sequence
 
 import torch
 
