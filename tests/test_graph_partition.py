@@ -145,3 +145,49 @@ def test_inference_partition_uses_forward_time():
         for i, (ft, bw) in enumerate([(1, 0), (1, 0), (1, 8), (1, 0)])])
     infer = partition_chain(g2, 2, straight=True, inference=True)
     assert infer.module_to_stage_map == [0, 0, 1, 1]
+
+
+def test_partition_chain_optimal_vs_bruteforce():
+    """DP result matches brute-force enumeration of every contiguous
+    partition + replication assignment on small random chains."""
+    import itertools
+    import random
+
+    from ddlbench_amd.parallel.pipeline.partition import (_dp_allreduce_time,
+                                                          partition_chain)
+    rng = random.Random(7)
+    BW = 100e9
+    for trial in range(12):
+        n = rng.randint(2, 6)
+        m = rng.randint(2, 4)
+        times = [rng.uniform(0.5, 4.0) for _ in range(n)]
+        acts = [rng.uniform(1e8, 1e10) for _ in range(n)]
+        params = [rng.uniform(1e6, 1e8) for _ in range(n)]
+        g = _chain(times, acts, params)
+        res = partition_chain(g, m)
+
+        def cost_of(cuts, reps):
+            # cuts: stage boundaries; reps: replicas per stage
+            cost = 0.0
+            for (a, b), r in zip(cuts, reps):
+                T = sum(times[a:b])
+                P = sum(params[a:b])
+                st = T / r + _dp_allreduce_time(r, P, BW)
+                ci = (2.0 * acts[a - 1] / (BW * r)) if a > 0 else 0.0
+                cost = max(cost, st, ci)
+            return cost
+
+        best = float("inf")
+        # enumerate all ways to split n layers into s stages and give
+        # each stage >=1 replica summing to exactly <= m ... the DP uses
+        # exactly m machines, so require sum(reps) == m
+        for s in range(1, min(n, m) + 1):
+            for cutpts in itertools.combinations(range(1, n), s - 1):
+                bounds = [0] + list(cutpts) + [n]
+                stages = list(zip(bounds[:-1], bounds[1:]))
+                for reps in itertools.product(range(1, m + 1), repeat=s):
+                    if sum(reps) != m:
+                        continue
+                    best = min(best, cost_of(stages, reps))
+        assert abs(res.bottleneck - best) < 1e-9 * max(best, 1), \
+            (trial, res.bottleneck, best)
