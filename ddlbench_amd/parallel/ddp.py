@@ -119,7 +119,14 @@ class BucketedDataParallel(nn.Module):
 
     def finalize_backward(self) -> None:
         """Wait for in-flight reductions and average. Call between
-        loss.backward() and optimizer.step()."""
+        loss.backward() and optimizer.step().
+
+        Contract: exactly one backward per zero_grad_buckets()/
+        finalize_backward() pair — a bucket's all-reduce launches when
+        its last grad lands, so a second accumulation pass would race
+        the in-flight collective (the reference's horovod path has the
+        same one-pass default, batches_per_allreduce=1,
+        imagenet_horovod.py:36)."""
         if self.world_size <= 1:
             return
         for b in self._buckets:

@@ -170,6 +170,24 @@ class Graph:
                     res.append((n.node_id, dst))
         return res
 
+    def to_dot(self) -> str:
+        """Graphviz text (the reference's to_dot, graph.py:482-616) —
+        node label = desc + compute ms, cluster color by stage."""
+        lines = ["digraph layers {", "  rankdir=TB;"]
+        for n in self.topological_sort():
+            ms = n.compute_time * 1e3
+            label = f"{n.node_id}: {n.desc}\\n{ms:.2f} ms"
+            color = f"/set312/{(n.stage_id % 12) + 1}" if n.stage_id >= 0 \
+                else "white"
+            lines.append(
+                f'  n{n.node_id} [label="{label}", style=filled, '
+                f'fillcolor="{color}"];')
+        for src in sorted(self.edges):
+            for dst in self.edges[src]:
+                lines.append(f"  n{src} -> n{dst};")
+        lines.append("}")
+        return "\n".join(lines) + "\n"
+
     # ---- serialization (our own line format) ----------------------------
     def dumps(self) -> str:
         lines = []

@@ -191,3 +191,13 @@ def test_partition_chain_optimal_vs_bruteforce():
                     best = min(best, cost_of(stages, reps))
         assert abs(res.bottleneck - best) < 1e-9 * max(best, 1), \
             (trial, res.bottleneck, best)
+
+
+def test_to_dot():
+    g = _chain([1.0, 2.0])
+    g.nodes[0].stage_id = 0
+    g.nodes[1].stage_id = 1
+    dot = g.to_dot()
+    assert dot.startswith("digraph")
+    assert "n0 -> n1;" in dot
+    assert "1000.00 ms" in dot  # _chain times are in seconds
