@@ -219,3 +219,69 @@ def test_gnmt_1f1b_pipeline_two_stages(free_port):
 @pytest.mark.timeout(600)
 def test_gnmt_1f1b_pipeline_three_stages(free_port):
     mp.spawn(_worker_gnmt_pipe, args=(3, free_port), nprocs=3, join=True)
+
+
+# ------------------------------------- deep pipeline: 4 stages, warmup 3
+def _worker_deep(rank, world, port):
+    _env(rank, world, port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from ddlbench_amd.ops.sgd import FusedSGD
+    from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
+    from ddlbench_amd.parallel.pipeline.runtime import (StagePlan,
+                                                        StageRuntime)
+    from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
+
+    torch.manual_seed(1)
+    layers = [torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Tanh())
+              for _ in range(4)]
+    layers.append(torch.nn.Linear(8, 3))
+    plan = StagePlan(replicas=[1, 1, 1, 1])
+    tr = PipelineTransport(plan.edges(), backend="gloo")
+    mods = [torch.nn.Sequential(layers[0]),
+            torch.nn.Sequential(layers[1]),
+            torch.nn.Sequential(layers[2]),
+            torch.nn.Sequential(layers[3], layers[4])]
+    mod = mods[rank]
+    B = 4
+    out_dim = 8 if rank < 3 else 3
+    rt = StageRuntime(plan, rank, mod, tr,
+                      in_shape=None if rank == 0 else torch.Size([B, 8]),
+                      out_shape=torch.Size([B, out_dim]),
+                      device=torch.device("cpu"), dtype=torch.float32,
+                      loss_fn=torch.nn.functional.cross_entropy)
+    warmup = plan.num_warmup(rank)
+    assert warmup == 3 - rank
+    opt = VersionedOptimizer(FusedSGD(mod.parameters(), lr=0.05,
+                                      momentum=0.9, backend="torch"),
+                             versioned=warmup > 0)
+    gen = torch.Generator().manual_seed(3)
+    N = 10
+    xs = [torch.randn(B, 8, generator=gen) for _ in range(N)]
+    ys = [torch.randint(3, (B,), generator=gen) for _ in range(N)]
+    mbs = rt.my_minibatches(N)
+    losses = []
+    for k in range(warmup):
+        loss, _ = rt.run_forward(mbs[k], lambda i: xs[i], lambda i: ys[i])
+        if loss is not None:
+            losses.append(loss.detach())
+    for k in range(len(mbs)):
+        if warmup + k < len(mbs):
+            loss, _ = rt.run_forward(mbs[warmup + k], lambda i: xs[i],
+                                     lambda i: ys[i])
+            if loss is not None:
+                losses.append(loss.detach())
+        opt.zero_grad(set_to_none=False)
+        rt.run_backward()
+        opt.step()
+    assert rt.stats.fwd_count == N and rt.stats.bwd_count == N
+    assert len(rt.inflight) == 0
+    if rank == 3:
+        assert len(losses) == N
+        assert all(torch.isfinite(l) for l in losses)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_deep_pipeline_four_stages(free_port):
+    mp.spawn(_worker_deep, args=(4, free_port), nprocs=4, join=True)
