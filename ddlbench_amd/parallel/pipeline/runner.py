@@ -89,8 +89,11 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
     for s in range(plan.num_stages):
         layers = [i for i, t in enumerate(m2s) if t == s]
         stage_slices.append(torch.nn.Sequential(*[seq[i] for i in layers]))
+    # profiling moved `seq` (and thus the slices) onto `device` on rank 0,
+    # so the probe must follow it
     probe = torch.zeros((2,) + tuple(cfg.shape), dtype=dtype)
-    shapes = dry_run_shapes(stage_slices, probe)
+    probe_dev = next(seq.parameters()).device
+    shapes = dry_run_shapes(stage_slices, probe, device=probe_dev)
     B = cfg.batch_size
 
     def with_batch(shape):
