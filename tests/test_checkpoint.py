@@ -55,3 +55,51 @@ def test_single_strategy_resume(tmp_path):
                        checkpoint_dir=str(tmp_path), resume=True)
     res = run_single(cfg2)  # epoch already done -> returns without training
     assert res["sec_per_epoch"] == 0.0
+
+
+def test_bnact_num_batches_tracked_syncs_on_state_dict():
+    from ddlbench_amd.ops.modules import BNAct
+    m = BNAct(4)
+    m.train()
+    x = torch.randn(2, 4, 3, 3)
+    m(x)
+    m(x)
+    sd = m.state_dict()
+    assert sd["num_batches_tracked"].item() == 2
+    m2 = BNAct(4)
+    m2.load_state_dict(sd)
+    assert m2.num_batches_tracked.item() == 2
+
+
+def test_mixed_dtype_param_groups_cpu():
+    """FusedSGD with bf16 + f32 params in one group (the BNAct-keeps-f32
+    situation under a bf16 model)."""
+    from ddlbench_amd.ops.sgd import FusedSGD
+    torch.manual_seed(0)
+    a = torch.nn.Parameter(torch.randn(4, 4, dtype=torch.bfloat16))
+    b = torch.nn.Parameter(torch.randn(4))
+    opt = FusedSGD([a, b], lr=0.1, momentum=0.9, backend="torch")
+    (a.float().sum() + b.sum()).backward()
+    pa, pb = a.detach().clone(), b.detach().clone()
+    opt.step()
+    assert not torch.equal(a, pa) and not torch.equal(b, pb)
+
+
+def test_versioned_optimizer_checkpoint_roundtrip(tmp_path):
+    """Stage checkpoint through the VersionedOptimizer wrapper."""
+    from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
+    torch.manual_seed(0)
+    m = torch.nn.Linear(4, 4)
+    opt = VersionedOptimizer(FusedSGD(m.parameters(), lr=0.1,
+                                      momentum=0.9, backend="torch"))
+    for _ in range(2):
+        opt.zero_grad(set_to_none=False)
+        m(torch.randn(2, 4)).sum().backward()
+        opt.step()
+    save_stage_checkpoint(str(tmp_path), 1, 5, "x", m, opt)
+    m2 = torch.nn.Linear(4, 4)
+    opt2 = VersionedOptimizer(FusedSGD(m2.parameters(), lr=0.1,
+                                       momentum=0.9, backend="torch"))
+    st = load_stage_checkpoint(str(tmp_path), 1, m2, opt2)
+    assert st["epoch"] == 5
+    torch.testing.assert_close(m2.weight, m.weight)
