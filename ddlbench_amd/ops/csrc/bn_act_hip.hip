@@ -404,13 +404,12 @@ __global__ void bn_apply_nhwc_kernel2(const T* __restrict__ x,
   const int RG = blockDim.x / CG8;
   const int64_t c0 = ((int64_t)blockIdx.x * CG8 + ci) * 8;
   if (c0 >= C || rj >= RG) return;
-  float sc[8], sh[8], rs[8];
+  float sc[8], sh[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     sc[j] = invstd[c0 + j] * gamma[c0 + j];
     sh[j] = beta[c0 + j] - mean[c0 + j] * sc[j];
   }
-  (void)rs;
   const int64_t per = (rows + gridDim.y - 1) / gridDim.y;
   const int64_t begin = (int64_t)blockIdx.y * per;
   const int64_t end = i64min(begin + per, rows);
