@@ -19,3 +19,16 @@ def test_run_sh_rejects_bad_framework():
         ["bash", "run/run.sh", "-f", "nope"],
         capture_output=True, text=True, timeout=60)
     assert out.returncode != 0
+
+
+def test_run_sh_horovod_two_ranks_cpu(tmp_path):
+    """The torchrun launch path of the harness (2 gloo ranks on CPU)."""
+    import os
+    env = dict(os.environ)
+    env.update(HOME=str(tmp_path), MASTER_PORT="29541")
+    out = subprocess.run(
+        ["bash", "run/run.sh", "-b", "mnist", "-f", "horovod", "-g", "2",
+         "-m", "resnet18", "-e", "1", "-B", "16", "-p", "0"],
+        capture_output=True, text=True, timeout=900, env=env)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-1000:]
+    assert "valid accuracy:" in out.stdout
