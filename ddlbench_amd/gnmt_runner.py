@@ -155,7 +155,8 @@ def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
                                           gnmt_pipeline_units)
     from ddlbench_amd.parallel import init_distributed
     from ddlbench_amd.parallel.pipeline.comm import PipelineTransport
-    from ddlbench_amd.parallel.pipeline.balance import partition_minmax
+    from ddlbench_amd.parallel.pipeline.balance import (partition_minmax,
+                                                        profile_unit_times)
     from ddlbench_amd.parallel.pipeline.runtime import (StagePlan,
                                                         StageRuntime)
     from ddlbench_amd.parallel.pipeline.stash import VersionedOptimizer
@@ -177,11 +178,40 @@ def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
     if dt != torch.float32:
         model = model.to(dt)
     units = gnmt_pipeline_units(model)
+    assert world <= len(units), \
+        f"at most {len(units)} pipeline stages for this GNMT config"
     specs = gnmt_edge_specs(model, len(units), src_len_max, tgt_len,
                             batch_size, dt)
 
-    # even split by unit count (decoder layers dominate roughly equally)
-    sizes = partition_minmax([1.0] * len(units), world)
+    # profiled stage balance: rank 0 times each unit on one minibatch
+    # shape and broadcasts the split (the 1F1B image runner's
+    # profile->partition->broadcast flow, here over tuple units)
+    if world > 1 and dist.is_initialized():
+        payload = [None]
+        if env.rank == 0:
+            if dev.type == "cuda":
+                for u in units:
+                    u.to(dev)  # profile on the real device
+            g0 = torch.Generator().manual_seed(seed)
+            sample = (torch.randint(3, vocab, (src_len_max, batch_size),
+                                    generator=g0),
+                      torch.full((batch_size,), src_len_max,
+                                 dtype=torch.long),
+                      torch.randint(3, vocab, (tgt_len, batch_size),
+                                    generator=g0))
+            times = profile_unit_times(units, sample)
+            payload = [partition_minmax(times, world)]
+        dist.broadcast_object_list(payload, src=0)
+        sizes = payload[0]
+    else:
+        sizes = partition_minmax([1.0] * len(units), world)
+    if env.rank == 0 and dev.type == "cuda":
+        # release the units that belong to other stages
+        b0 = sum(sizes[:env.rank])
+        for i, u in enumerate(units):
+            if not (b0 <= i < b0 + sizes[env.rank]):
+                u.to("cpu")
+        torch.cuda.empty_cache()
     bounds = [0]
     for sz in sizes:
         bounds.append(bounds[-1] + sz)

@@ -41,6 +41,35 @@ def profile_module_times(seq: nn.Sequential, sample: torch.Tensor,
     return [t / iters for t in times]
 
 
+@torch.no_grad()
+def profile_unit_times(units, sample_args, iters: int = 3,
+                       warmup: int = 1) -> List[float]:
+    """Per-unit forward time for a tuple-I/O unit chain (the GNMT
+    pipeline units). CPU/GPU device follows the units' parameters."""
+    times = [0.0] * len(units)
+    dev = None
+    for u in units:
+        for prm in u.parameters():
+            dev = prm.device
+            break
+        if dev is not None:
+            break
+    dev = dev or torch.device("cpu")
+    for it in range(warmup + iters):
+        xs = tuple(t.to(dev) for t in sample_args)
+        for i, u in enumerate(units):
+            if dev.type == "cuda":
+                torch.cuda.synchronize(dev)
+            t0 = time.perf_counter()
+            out = u(*xs)
+            if dev.type == "cuda":
+                torch.cuda.synchronize(dev)
+            if it >= warmup:
+                times[i] += time.perf_counter() - t0
+            xs = (out,) if torch.is_tensor(out) else tuple(out)
+    return [t / iters for t in times]
+
+
 def partition_minmax(weights: Sequence[float], k: int) -> List[int]:
     """Split weights into k contiguous groups minimizing the max group
     sum. Returns group sizes (len k, sums to len(weights)). Exact DP."""
