@@ -49,9 +49,11 @@ def _make_plan(cfg: BenchConfig, seq, device, world: int):
         graph = profile_sequential(seq, sample, device=device,
                                    iters=3, warmup=1)
         result = partition_chain(graph, world, straight=straight)
-        os.makedirs(f"profiles/{cfg.arch}", exist_ok=True)
-        graph.save(f"profiles/{cfg.arch}/graph.txt")
-        result.save(f"profiles/{cfg.arch}/conf.json")
+        prof_dir = os.path.join(
+            os.environ.get("DDLB_PROFILE_DIR", "profiles"), cfg.arch)
+        os.makedirs(prof_dir, exist_ok=True)
+        graph.save(os.path.join(prof_dir, "graph.txt"))
+        result.save(os.path.join(prof_dir, "conf.json"))
         payload = [{
             "module_to_stage_map": result.module_to_stage_map,
             "replicas": [s.replicas for s in result.stages],

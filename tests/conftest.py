@@ -1,5 +1,13 @@
+import os
+import tempfile
+
 import pytest
 import torch
+
+# keep test-generated profiler artifacts out of the committed profiles/
+os.environ.setdefault("DDLB_PROFILE_DIR",
+                      os.path.join(tempfile.gettempdir(),
+                                   "ddlb_test_profiles"))
 
 
 def pytest_configure(config):
