@@ -15,10 +15,15 @@ from ddlbench_amd.config import DATASET_SHAPES
 from ddlbench_amd.models.resnet import ResNet
 from ddlbench_amd.models.vgg import VGG
 from ddlbench_amd.models.mobilenetv2 import MobileNetV2
+from ddlbench_amd.models.extra import (DenseNet, Inception3, MobileNetV1,
+                                       SqueezeNet)
 
-RESNETS = ("resnet18", "resnet34", "resnet50", "resnet101", "resnet152")
+RESNETS = ("resnet18", "resnet34", "resnet50", "resnet101", "resnet152",
+           "resnext50_32x4d")
 VGGS = ("vgg11", "vgg13", "vgg16", "vgg19")
-ARCHS = RESNETS + VGGS + ("mobilenetv2",)
+DENSENETS = ("densenet121", "densenet169")
+ARCHS = RESNETS + VGGS + DENSENETS + (
+    "mobilenetv2", "mobilenetv1", "squeezenet", "inception3")
 
 
 def build_model(dataset: str, arch: str) -> nn.Module:
@@ -28,8 +33,19 @@ def build_model(dataset: str, arch: str) -> nn.Module:
         return ResNet(arch, in_channels=c, num_classes=ncls, stem=stem)
     if arch in VGGS:
         return VGG(arch, in_channels=c, num_classes=ncls, stem=stem)
+    if arch in DENSENETS:
+        return DenseNet(arch, in_channels=c, num_classes=ncls, stem=stem)
     if arch == "mobilenetv2":
         return MobileNetV2(in_channels=c, num_classes=ncls, stem=stem)
+    if arch == "mobilenetv1":
+        return MobileNetV1(in_channels=c, num_classes=ncls, stem=stem)
+    if arch == "squeezenet":
+        return SqueezeNet(in_channels=c, num_classes=ncls, stem=stem)
+    if arch == "inception3":
+        if dataset not in ("imagenet", "highres"):
+            raise ValueError("inception3 needs imagenet/highres inputs "
+                             "(aggressive stem downsampling)")
+        return Inception3(in_channels=c, num_classes=ncls)
     raise ValueError(f"unknown arch {arch!r}; choose from {ARCHS}")
 
 

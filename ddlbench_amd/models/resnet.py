@@ -19,8 +19,9 @@ import torch.nn.functional as F
 from ddlbench_amd.ops.modules import BNAct
 
 
-def conv3x3(cin, cout, stride=1):
-    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+def conv3x3(cin, cout, stride=1, groups=1):
+    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False,
+                     groups=groups)
 
 
 def conv1x1(cin, cout, stride=1):
@@ -57,15 +58,22 @@ class BasicBlock(nn.Module):
 
 
 class Bottleneck(nn.Module):
+    """1x1 -> 3x3 -> 1x1 with fused BN/ReLU/residual. groups/base_width
+    give the ResNeXt variant (aggregated transforms — the reference's
+    profiler-zoo resnext)."""
+
     expansion = 4
+    groups = 1
+    base_width = 64
 
     def __init__(self, cin, planes, stride=1):
         super().__init__()
-        self.conv1 = conv1x1(cin, planes)
-        self.bn1 = BNAct(planes, act="relu")
-        self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = BNAct(planes, act="relu")
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        width = int(planes * (self.base_width / 64.0)) * self.groups
+        self.conv1 = conv1x1(cin, width)
+        self.bn1 = BNAct(width, act="relu")
+        self.conv2 = conv3x3(width, width, stride, groups=self.groups)
+        self.bn2 = BNAct(width, act="relu")
+        self.conv3 = conv1x1(width, planes * self.expansion)
         self.bn3 = BNAct(planes * self.expansion, act="relu")
         self.downsample = None
         if stride != 1 or cin != planes * self.expansion:
@@ -76,6 +84,11 @@ class Bottleneck(nn.Module):
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         return self.bn3(self.conv3(out), res=identity)
+
+
+class BottleneckX(Bottleneck):
+    groups = 32
+    base_width = 4
 
 
 class Stem(nn.Module):
@@ -114,6 +127,7 @@ _LAYERS = {
     "resnet50": (Bottleneck, (3, 4, 6, 3)),
     "resnet101": (Bottleneck, (3, 4, 23, 3)),
     "resnet152": (Bottleneck, (3, 8, 36, 3)),
+    "resnext50_32x4d": (BottleneckX, (3, 4, 6, 3)),
 }
 
 

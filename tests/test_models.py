@@ -44,7 +44,8 @@ def test_sequential_matches_model(dataset, arch):
 
 def test_all_archs_instantiate():
     for arch in ARCHS:
-        m = build_model("cifar10", arch)
+        ds = "imagenet" if arch == "inception3" else "cifar10"
+        m = build_model(ds, arch)
         assert sum(p.numel() for p in m.parameters()) > 0
 
 
@@ -60,3 +61,57 @@ def test_highres_shape_forward():
     with torch.no_grad():
         y = m(torch.randn(1, 3, 512, 512))
     assert y.shape == (1, 1000)
+
+
+EXTRA_CASES = [
+    ("imagenet", "densenet121"), ("imagenet", "squeezenet"),
+    ("imagenet", "resnext50_32x4d"), ("imagenet", "mobilenetv1"),
+    ("cifar10", "densenet121"), ("cifar10", "squeezenet"),
+    ("mnist", "mobilenetv1"),
+]
+
+
+@pytest.mark.parametrize("dataset,arch", EXTRA_CASES)
+def test_extended_zoo_forward_backward(dataset, arch):
+    torch.manual_seed(0)
+    c, h, w, ncls, _, _ = DATASET_SHAPES[dataset]
+    m = build_model(dataset, arch)
+    y = m(torch.randn(2, c, h, w))
+    assert y.shape == (2, ncls)
+    y.sum().backward()
+    assert all(p.grad is not None for p in m.parameters()
+               if p.requires_grad)
+
+
+def test_inception3_forward_backward():
+    m = build_model("imagenet", "inception3")
+    y = m(torch.randn(1, 3, 224, 224))
+    assert y.shape == (1, 1000)
+    y.sum().backward()
+
+
+def test_extended_zoo_param_count_parity():
+    """Exact parity with the torchvision/paper parameter counts."""
+    assert sum(p.numel() for p in
+               build_model("imagenet", "resnext50_32x4d").parameters()) \
+        == 25_028_904
+    assert sum(p.numel() for p in
+               build_model("imagenet", "densenet121").parameters()) \
+        == 7_978_856
+    assert sum(p.numel() for p in
+               build_model("imagenet", "squeezenet").parameters()) \
+        == 1_235_496
+
+
+@pytest.mark.parametrize("dataset,arch",
+                         [("cifar10", "densenet121"),
+                          ("cifar10", "squeezenet"),
+                          ("mnist", "mobilenetv1")])
+def test_extended_zoo_sequential_matches(dataset, arch):
+    torch.manual_seed(0)
+    c, h, w, ncls, _, _ = DATASET_SHAPES[dataset]
+    m = build_model(dataset, arch).eval()
+    seq = m.to_sequential().eval()
+    x = torch.randn(2, c, h, w)
+    with torch.no_grad():
+        torch.testing.assert_close(m(x), seq(x), rtol=1e-5, atol=1e-5)
