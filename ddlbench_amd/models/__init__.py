@@ -17,13 +17,15 @@ from ddlbench_amd.models.vgg import VGG
 from ddlbench_amd.models.mobilenetv2 import MobileNetV2
 from ddlbench_amd.models.extra import (DenseNet, Inception3, MobileNetV1,
                                        SqueezeNet)
+from ddlbench_amd.models.nasnet import NASNetAMobile
 
 RESNETS = ("resnet18", "resnet34", "resnet50", "resnet101", "resnet152",
            "resnext50_32x4d")
 VGGS = ("vgg11", "vgg13", "vgg16", "vgg19")
 DENSENETS = ("densenet121", "densenet169")
 ARCHS = RESNETS + VGGS + DENSENETS + (
-    "mobilenetv2", "mobilenetv1", "squeezenet", "inception3")
+    "mobilenetv2", "mobilenetv1", "squeezenet", "inception3",
+    "nasnetamobile")
 
 
 def build_model(dataset: str, arch: str) -> nn.Module:
@@ -41,6 +43,8 @@ def build_model(dataset: str, arch: str) -> nn.Module:
         return MobileNetV1(in_channels=c, num_classes=ncls, stem=stem)
     if arch == "squeezenet":
         return SqueezeNet(in_channels=c, num_classes=ncls, stem=stem)
+    if arch == "nasnetamobile":
+        return NASNetAMobile(in_channels=c, num_classes=ncls, stem=stem)
     if arch == "inception3":
         if dataset not in ("imagenet", "highres"):
             raise ValueError("inception3 needs imagenet/highres inputs "

@@ -115,3 +115,19 @@ def test_extended_zoo_sequential_matches(dataset, arch):
     x = torch.randn(2, c, h, w)
     with torch.no_grad():
         torch.testing.assert_close(m(x), seq(x), rtol=1e-5, atol=1e-5)
+
+
+def test_nasnetamobile():
+    torch.manual_seed(0)
+    for ds in ("imagenet", "cifar10"):
+        c, h, w, ncls, _, _ = DATASET_SHAPES[ds]
+        m = build_model(ds, "nasnetamobile")
+        y = m(torch.randn(1, c, h, w))
+        assert y.shape == (1, ncls)
+        y.sum().backward()
+    # tuple-passing sequential flattening matches the model
+    m = build_model("cifar10", "nasnetamobile").eval()
+    seq = m.to_sequential().eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        torch.testing.assert_close(m(x), seq(x), rtol=1e-5, atol=1e-5)
