@@ -20,7 +20,19 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 ROOT = Path(__file__).resolve().parent
 CSRC = ROOT / "ddlbench_amd" / "ops" / "csrc"
 
-sources = sorted(str(p) for p in CSRC.glob("*.hip"))
+# Hand-written kernel sources, pinned explicitly (torch's hipify step
+# generates *_hip.hip siblings at build time; those are never tracked).
+KERNEL_SOURCES = [
+    "bindings.hip",
+    "bn_act.hip",
+    "conv_mfma.hip",
+    "conv_wgrad.hip",
+    "cross_entropy.hip",
+    "depthwise_conv.hip",
+    "fused_sgd.hip",
+    "seq_utils.hip",
+]
+sources = [str(CSRC / s) for s in KERNEL_SOURCES]
 
 setup(
     name="ddlbench_amd_ops",
