@@ -57,6 +57,8 @@ class BenchConfig:
     synthetic_scale: float = 1.0      # fraction of the full dataset size to use
     data_dir: str = ""
     lr: float = 0.01
+    lr_schedule: str = "constant"     # constant | step30 | warmup (utils/lr.py)
+    warmup_epochs: int = 5            # warmup schedule ramp length
     momentum: float = 0.9
     weight_decay: float = 0.0
     dtype: str = "float32"            # compute dtype: float32 | bfloat16
@@ -128,6 +130,11 @@ def make_parser(default_arch: str = "resnet18") -> argparse.ArgumentParser:
                         "for smoke runs). -1 selects the highres variant where "
                         "supported.")
     p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--lr-schedule", default="constant",
+                   choices=["constant", "step30", "warmup"],
+                   help="step30 = imagenet /30-epoch decay; warmup = "
+                        "horovod 5-epoch ramp to lr*N then 30/60/80 steps")
+    p.add_argument("--warmup-epochs", type=int, default=5)
     p.add_argument("--momentum", type=float, default=0.9)
     p.add_argument("--weight-decay", type=float, default=0.0)
     p.add_argument("--dtype", default="float32", choices=["float32", "bfloat16"])
@@ -146,6 +153,8 @@ def config_from_args(dataset: str, strategy: str, args: argparse.Namespace,
         arch=args.arch,
         synthetic_scale=(1.0 if highres else max(args.synthetic_scale, 0.0) or 1.0),
         lr=args.lr,
+        lr_schedule=getattr(args, "lr_schedule", "constant"),
+        warmup_epochs=getattr(args, "warmup_epochs", 5),
         momentum=args.momentum,
         weight_decay=args.weight_decay,
         dtype=args.dtype,

@@ -70,6 +70,15 @@ class Trainer:
         self.dtype = compute_dtype(cfg)
         # set when the model is wrapped in BucketedDataParallel
         self.dp = model if hasattr(model, "finalize_backward") else None
+        # LR schedule over the optimizer's configured base LR (already
+        # x world_size for DDP) — reference imagenet_pytorch.py:225-229,
+        # imagenet_horovod.py:258-275
+        from ddlbench_amd.utils.lr import make_lr_schedule
+        self.lr_schedule = make_lr_schedule(
+            cfg.lr_schedule, world_size=world_size,
+            warmup_epochs=cfg.warmup_epochs)
+        self._base_lrs = [g["lr"] for g in optimizer.param_groups]
+        self._warmup_active = cfg.lr_schedule == "warmup"
 
     def _to_device(self, x: torch.Tensor, y: torch.Tensor):
         x = x.to(self.device, dtype=self.dtype, non_blocking=True)
@@ -81,15 +90,28 @@ class Trainer:
     def train_epoch(self, loader, epoch: int) -> tuple:
         self.model.train()
         cfg = self.cfg
-        losses = AverageMeter()
         n_batches = len(loader)
         seen = 0
+        # loss accumulates ON DEVICE; .item() only at log points so the
+        # steady-state loop never host-syncs (the reference's per-batch
+        # loss read under-reports samples/sec; cf. runner.py deferral)
+        loss_sum = None
         if self.device.type == "cuda":
             torch.cuda.reset_peak_memory_stats(self.device)
+        from ddlbench_amd.utils.lr import apply_lr
+        apply_lr(self.optimizer, self._base_lrs,
+                 self.lr_schedule(epoch, 0.0))
+        # the warmup ramp is per-batch in the reference
+        # (imagenet_horovod.py:262: epoch + (batch+1)/len(loader))
+        per_batch_lr = (self._warmup_active
+                        and epoch <= self.cfg.warmup_epochs)
         sync(self.device)
         tick = time.perf_counter()
         window_start, window_samples = tick, 0
         for i, (x, y) in enumerate(loader):
+            if per_batch_lr:
+                apply_lr(self.optimizer, self._base_lrs,
+                         self.lr_schedule(epoch, (i + 1) / n_batches))
             x, y = self._to_device(x, y)
             out = self.model(x)
             loss = self.loss_fn(out, y)
@@ -101,7 +123,10 @@ class Trainer:
                 self.optimizer.zero_grad(set_to_none=True)
                 loss.backward()
             self.optimizer.step()
-            losses.update(loss.item(), y.size(0))
+            with torch.no_grad():
+                contrib = loss.detach() * y.size(0)
+                loss_sum = contrib if loss_sum is None \
+                    else loss_sum.add_(contrib)
             seen += y.size(0)
             window_samples += y.size(0)
             if cfg.log_interval and (i + 1) % cfg.log_interval == 0:
@@ -116,7 +141,8 @@ class Trainer:
         sync(self.device)
         elapsed = time.perf_counter() - tick
         sps = seen * self.world_size / elapsed
-        return losses.avg, sps, elapsed
+        avg_loss = (float(loss_sum.item()) / seen) if seen else 0.0
+        return avg_loss, sps, elapsed
 
     @torch.no_grad()
     def validate(self, loader) -> tuple:

@@ -83,7 +83,12 @@ class FusedSGD(Optimizer):
             key = (gi, dtype)
             cached = self._cache.get(key)
             first_step = False
-            if cached is None or len(cached["params"]) != len(ps):
+            # invalidate on any identity change, not just count: a same-
+            # length set with different members (freeze/unfreeze, grads
+            # becoming None) would otherwise reuse stale device pointers
+            if (cached is None or len(cached["params"]) != len(ps)
+                    or any(a is not b
+                           for a, b in zip(cached["params"], ps))):
                 numels = [p.numel() for p in ps]
                 prefix = torch.zeros(len(ps), dtype=torch.int64)
                 total = 0
@@ -110,19 +115,26 @@ class FusedSGD(Optimizer):
                 self._cache[key] = cached
                 cached["first_step"] = first_step
             first_step = cached.pop("first_step", False)
+            # the fused kernel walks param/grad as flat buffers in
+            # storage order — any dense layout works (NCHW or
+            # channels_last) as long as grad strides match param's.
+            # On mismatch, hand the kernel a layout-matched COPY without
+            # rebinding p.grad: a rebind would orphan persistent DP
+            # bucket views (grads would silently stop reaching the
+            # all-reduce payload).
+            grads = []
             for p in ps:
-                # the fused kernel walks param/grad as flat buffers in
-                # storage order — any dense layout works (NCHW or
-                # channels_last) as long as grad strides match param's
-                if not _same_dense_layout(p, p.grad):
-                    p.grad = p.grad.contiguous() if p.is_contiguous() \
-                        else p.grad.contiguous(
-                            memory_format=torch.channels_last)
-                assert _same_dense_layout(p, p.grad), \
-                    "FusedSGD needs grads with the param's layout"
+                g = p.grad
+                if not _same_dense_layout(p, g):
+                    g = (g.contiguous() if p.is_contiguous()
+                         else g.contiguous(
+                             memory_format=torch.channels_last))
+                    assert _same_dense_layout(p, g), \
+                        "FusedSGD needs grads with the param's layout"
+                grads.append(g)
             # grads in persistent buckets (DP) keep their pointers —
             # re-upload the pointer array only when one moved
-            gptrs = [p.grad.data_ptr() for p in ps]
+            gptrs = [g.data_ptr() for g in grads]
             if cached.get("grad_ptrs_host") == gptrs:
                 ptr_grads = cached["ptr_grads"]
             else:

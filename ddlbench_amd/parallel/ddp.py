@@ -92,6 +92,21 @@ class BucketedDataParallel(nn.Module):
             h = p.register_post_accumulate_grad_hook(self._grad_ready)
             self._hooks.append(h)
 
+    @staticmethod
+    def _layout_view(flat_slice: torch.Tensor, p: torch.Tensor):
+        """View a flat bucket slice with the param's memory format.
+
+        A channels_last param needs a channels_last-strided grad view:
+        a row-major view_as() would make autograd accumulate NCHW-order
+        data while FusedSGD walks the param NHWC-order — and FusedSGD's
+        layout remediation would rebind p.grad to a fresh tensor,
+        orphaning the bucket (grads then never reach the all-reduce)."""
+        if (p.dim() == 4 and not p.is_contiguous()
+                and p.is_contiguous(memory_format=torch.channels_last)):
+            n, c, h, w = p.shape
+            return flat_slice.view(n, h, w, c).permute(0, 3, 1, 2)
+        return flat_slice.view_as(p)
+
     def _finish_bucket(self, bucket: _Bucket, dtype) -> None:
         device = bucket.params[0].device
         total = sum(p.numel() for p in bucket.params)
@@ -99,7 +114,7 @@ class BucketedDataParallel(nn.Module):
         off = 0
         for p in bucket.params:
             n = p.numel()
-            view = bucket.flat[off:off + n].view_as(p)
+            view = self._layout_view(bucket.flat[off:off + n], p)
             bucket.views[p] = view
             # persistent .grad view: accumulation writes into the payload
             p.grad = view

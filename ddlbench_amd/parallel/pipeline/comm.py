@@ -80,6 +80,12 @@ def dry_run_shapes(stages: List[torch.nn.Module], sample: torch.Tensor,
     with torch.no_grad():
         x = sample.to(device)
         for st in stages:
+            # a rank may have moved only ITS stage's layers to the GPU
+            # (they are shared with the full chain) — follow each
+            # slice's parameters so the probe never crosses devices
+            p = next(st.parameters(), None)
+            if p is not None and x.device != p.device:
+                x = x.to(p.device)
             x = st(x)
             shapes.append(x.shape)
     return shapes

@@ -282,10 +282,18 @@ def run_1f1b_training(cfg: BenchConfig) -> dict:
         if state is not None:
             start_epoch = state["epoch"] + 1
 
+    # per-epoch LR adjustment (reference main_with_runtime.py:441
+    # adjust_learning_rate; applies over the stage-scaled base LR)
+    from ddlbench_amd.utils.lr import apply_lr, make_lr_schedule
+    lr_sched = make_lr_schedule(cfg.lr_schedule, world_size=world,
+                                warmup_epochs=cfg.warmup_epochs)
+    base_lrs = [g["lr"] for g in opt.inner.param_groups]
+
     epoch_sps, epoch_secs = [], []
     val_loss = val_acc = 0.0
     train_loss = 0.0
     for epoch in range(start_epoch, cfg.epochs + 1):
+        apply_lr(opt.inner, base_lrs, lr_sched(epoch, 0.0))
         train_loss, sps, secs = train_epoch(epoch)
         val_loss, val_acc = validate()
         epoch_sps.append(sps)

@@ -74,6 +74,13 @@ def run_ddp(cfg: BenchConfig) -> dict:
     dp = BucketedDataParallel(model)
     # reference scales LR by world size (mnist_horovod.py:226)
     optimizer = make_optimizer(cfg, model, lr_scale=env.world_size)
+    start_epoch = 1
+    if cfg.resume and cfg.checkpoint_dir:
+        from ddlbench_amd.utils.checkpoint import load_stage_checkpoint
+        state = load_stage_checkpoint(cfg.checkpoint_dir, 0, model,
+                                      optimizer)
+        if state is not None:
+            start_epoch = state["epoch"] + 1
     train_loader, test_loader, sampler = make_loaders(
         cfg, world_size=env.world_size, rank=env.rank,
         pin_memory=device.type == "cuda")
@@ -82,7 +89,17 @@ def run_ddp(cfg: BenchConfig) -> dict:
                       world_size=env.world_size,
                       allreduce_metrics=lambda v: allreduce_mean_scalar(
                           v, device if device.type == "cuda" else None))
-    return trainer.fit(train_loader, test_loader, sampler)
+
+    def on_epoch_end(epoch, metrics):
+        # rank 0 only (reference: rank_in_stage==0 saves)
+        if cfg.checkpoint_dir and env.rank == 0:
+            from ddlbench_amd.utils.checkpoint import save_stage_checkpoint
+            save_stage_checkpoint(cfg.checkpoint_dir, 0, epoch, cfg.arch,
+                                  model, optimizer,
+                                  metrics.get("valid_accuracy", 0.0))
+
+    return trainer.fit(train_loader, test_loader, sampler,
+                       start_epoch=start_epoch, on_epoch_end=on_epoch_end)
 
 
 def run_gpipe(cfg: BenchConfig) -> dict:

@@ -28,7 +28,9 @@ def save_stage_checkpoint(ckpt_dir: str, stage: int, epoch: int, arch: str,
         "arch": arch,
         "stage": stage,
         "state_dict": module.state_dict(),
-        "best_metric": best_metric,
+        # reference key name (main_with_runtime.py:393-403) so
+        # checkpoints stay cross-readable with reference tooling
+        "best_prec1": best_metric,
     }
     if optimizer is not None:
         inner = getattr(optimizer, "inner", optimizer)
@@ -48,6 +50,8 @@ def load_stage_checkpoint(ckpt_dir: str, stage: int,
     if not os.path.exists(path):
         return None
     state = torch.load(path, map_location=map_location, weights_only=False)
+    if "best_prec1" not in state:  # round-1 checkpoints used best_metric
+        state["best_prec1"] = state.get("best_metric", 0.0)
     module.load_state_dict(state["state_dict"])
     if optimizer is not None and "optimizer" in state:
         inner = getattr(optimizer, "inner", optimizer)
