@@ -68,7 +68,13 @@ def main():
         ext = require_extension()
         res = {"shape": f"C{C}_HW{HW}_K{K}_R{R}_s{stride}",
                "batch": N}
-        # forward
+        # forward: v2 (deep pipeline, default), v1 (128-tile), MIOpen
+        os.environ["DDLB_CONV_V2"] = "1"
+        t = bench_op(lambda: ext.conv_igemm_fwd(x, w, stride, pad),
+                     args.iters, args.warmup)
+        res["mfma2_fwd_ms"] = round(t * 1e3, 3)
+        res["mfma2_fwd_tf"] = round(flops / t / 1e12, 1)
+        os.environ["DDLB_CONV_V2"] = "0"
         t = bench_op(lambda: ext.conv_igemm_fwd(x, w, stride, pad),
                      args.iters, args.warmup)
         res["mfma_fwd_ms"] = round(t * 1e3, 3)
@@ -78,11 +84,19 @@ def main():
         res["miopen_fwd_ms"] = round(t * 1e3, 3)
         res["miopen_fwd_tf"] = round(flops / t / 1e12, 1)
         # dgrad
+        os.environ["DDLB_CONV_V2"] = "1"
+        t = bench_op(lambda: ext.conv_igemm_dgrad(
+            dy, w_perm, N, C, HW, HW, stride, pad),
+            args.iters, args.warmup)
+        res["mfma2_dgrad_ms"] = round(t * 1e3, 3)
+        res["mfma2_dgrad_tf"] = round(flops / t / 1e12, 1)
+        os.environ["DDLB_CONV_V2"] = "0"
         t = bench_op(lambda: ext.conv_igemm_dgrad(
             dy, w_perm, N, C, HW, HW, stride, pad),
             args.iters, args.warmup)
         res["mfma_dgrad_ms"] = round(t * 1e3, 3)
         res["mfma_dgrad_tf"] = round(flops / t / 1e12, 1)
+        os.environ["DDLB_CONV_V2"] = "1"
         t = bench_op(lambda: torch.ops.aten.convolution_backward(
             dy, x, w, None, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1, [True, False, False]), args.iters, args.warmup)

@@ -25,12 +25,13 @@
 // dilation=1, C %8 == 0 (FPROP) / K %8 == 0 (DGRAD) so no 16-B chunk
 // crosses an (r,s) boundary.
 
-#include "common.h"
+#include "conv_igemm.h"
 #include <stdint.h>
 #include <stdexcept>
+#include <stdlib.h>
 #include <string>
 
-using bf16 = __hip_bfloat16;
+using bf16 = conv_bf16;
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
@@ -39,20 +40,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define BK 64
 #define THREADS 256
 
-struct ConvParams {
-  const bf16* a;      // x (FPROP) or dy (DGRAD), NHWC memory
-  const bf16* b;      // w as (K,R,S,C) mem (FPROP) or (C,R,S,K) (DGRAD)
-  bf16* out;          // y (FPROP: M x K) or dx (DGRAD: M x C)
-  const bf16* zero;   // >=16B of zeros
-  int N, H, W, Cin;   // logical input dims of the *forward* conv
-  int K, OH, OW;      // output channels / spatial of the forward conv
-  int R, S, stride, pad;
-  long M, Nd, Kd;     // GEMM dims of THIS pass
-  // stride-2 dgrad parity class (MODE 2): output pixels with
-  // ih%2==cls_a, iw%2==cls_b; valid taps r = r0+2*i (nr of them),
-  // s = s0+2*j (ns of them); nh/nw = pixel counts of the class
-  int cls_a, cls_b, r0, s0, nr, ns, nh, nw;
-};
+using RowCoords = ConvRowCoords;
 
 // ---- chunk -> global address generators -------------------------------
 // A-chunk: logical (row=m in [0,BM), cg in [0,8)) of the current K-step.
@@ -63,36 +51,11 @@ struct ConvParams {
 //
 // The A operand's row -> pixel decomposition is loop-invariant (the
 // GEMM row never changes across K-steps), so it is computed once per
-// staging slot (a_row_coords) and only the (r, s, c/k) part runs per
-// step (a_step_addr).
-struct RowCoords { int n, y, x; bool valid; };
-
+// staging slot (conv_a_row_coords) and only the (r, s, c/k) part runs
+// per step (a_step_addr).
 template <int MODE>
 DEV RowCoords a_row_coords(const ConvParams& p, long m) {
-  RowCoords rc;
-  rc.valid = m < p.M;
-  const long mm = rc.valid ? m : 0;
-  if (MODE == 0) {
-    const int ohw = p.OH * p.OW;
-    rc.n = (int)(mm / ohw);
-    const int rem = (int)(mm - (long)rc.n * ohw);
-    rc.y = rem / p.OW;
-    rc.x = rem - rc.y * p.OW;
-  } else if (MODE == 1) {
-    const int hw = p.H * p.W;
-    rc.n = (int)(mm / hw);
-    const int rem = (int)(mm - (long)rc.n * hw);
-    rc.y = rem / p.W;
-    rc.x = rem - rc.y * p.W;
-  } else {
-    const int hw = p.nh * p.nw;
-    rc.n = (int)(mm / hw);
-    const int rem = (int)(mm - (long)rc.n * hw);
-    const int ii = rem / p.nw, jj = rem - (rem / p.nw) * p.nw;
-    rc.y = 2 * ii + p.cls_a;
-    rc.x = 2 * jj + p.cls_b;
-  }
-  return rc;
+  return conv_a_row_coords<MODE>(p, m);
 }
 
 template <int MODE>
@@ -347,8 +310,14 @@ void launch_conv_igemm(const void* a, const void* b, void* out,
     else LAUNCH(MODE, 128, 128, 2, 2);                                      \
   } while (0)
 
+  // deep-pipeline structure (conv_mfma2.hip) first; DDLB_CONV_V2=0
+  // forces the 128-tile structure (re-read per call so benchmarks can
+  // A/B the two structures in one process)
+  const char* v2e = getenv("DDLB_CONV_V2");
+  const bool use_v2 = !(v2e && v2e[0] == '0');
+
   if (!dgrad) {
-    LAUNCH_TILED(0);
+    if (!(use_v2 && launch_conv_igemm_v2(p, 0, stream))) LAUNCH_TILED(0);
   } else if (stride == 2 && K % 8 == 0) {
     // four parity classes, each a dense reduction over its valid taps
     for (int a_ = 0; a_ < 2; ++a_) {
@@ -365,11 +334,12 @@ void launch_conv_igemm(const void* a, const void* b, void* out,
         if (p.nr <= 0 || p.ns <= 0 || p.nh <= 0 || p.nw <= 0) continue;
         p.M = (long)N * p.nh * p.nw;
         p.Kd = (long)p.nr * p.ns * K;
-        LAUNCH_TILED(2);
+        if (!(use_v2 && launch_conv_igemm_v2(p, 2, stream)))
+          LAUNCH_TILED(2);
       }
     }
   } else {
-    LAUNCH_TILED(1);
+    if (!(use_v2 && launch_conv_igemm_v2(p, 1, stream))) LAUNCH_TILED(1);
   }
 #undef LAUNCH_TILED
 #undef LAUNCH

@@ -26,6 +26,7 @@ KERNEL_SOURCES = [
     "bindings.hip",
     "bn_act.hip",
     "conv_mfma.hip",
+    "conv_mfma2.hip",
     "conv_wgrad.hip",
     "cross_entropy.hip",
     "depthwise_conv.hip",

@@ -216,6 +216,12 @@ def test_depthwise_conv(dtype, tol, stride):
     (4, 64, 14, 14, 128, 1, 2, 0),      # 1x1 s2 (degenerate dgrad classes)
     (2, 128, 7, 7, 120, 3, 1, 1),       # Nd not multiple of tile
     (2, 16, 9, 9, 24, 3, 2, 1),         # small C/K (mobilenet-ish)
+    # deep-pipeline (v2) coverage: Nd >= 96 both directions, M/Kd tails,
+    # multi-column-block Nd, stride-2 parity classes at v2 widths
+    (3, 96, 13, 13, 104, 3, 1, 1),      # v2 both dirs, every tail odd
+    (2, 128, 9, 9, 512, 1, 1, 0),       # v2 fwd 4 column blocks
+    (2, 256, 9, 9, 96, 3, 2, 1),        # v2 dgrad parity classes
+    (2, 96, 16, 16, 96, 1, 1, 0),       # v2, Kd=96 (not mult of 64)
 ])
 def test_conv_mfma_fwd_dgrad(shape):
     from ddlbench_amd.ops.conv import conv2d_mfma
@@ -398,3 +404,43 @@ def test_model_trains_on_native_kernels():
         assert losses[-1] < losses[0]
     finally:
         set_default_backend("auto")
+
+
+@pytest.mark.parametrize("shape", [
+    (3, 96, 13, 13, 104, 3, 1, 1),
+    (2, 128, 9, 9, 512, 1, 1, 0),
+    (2, 256, 10, 10, 128, 3, 2, 1),
+])
+def test_conv_v2_matches_v1(shape):
+    """The deep-pipeline (v2) and 128-tile (v1) structures accumulate
+    the identical K-ordered f32 MFMA chain -> bitwise-equal outputs."""
+    import os
+    from ddlbench_amd.ops import require_extension
+    ext = require_extension()
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(0)
+    dev = _dev()
+    cl = torch.channels_last
+    x = torch.randn(N, C, H, W, device=dev,
+                    dtype=torch.bfloat16).contiguous(memory_format=cl)
+    w = torch.randn(K, C, R, R, device=dev,
+                    dtype=torch.bfloat16).contiguous(memory_format=cl)
+    OH = (H + 2 * pad - R) // stride + 1
+    dy = torch.randn(N, K, OH, OH, device=dev,
+                     dtype=torch.bfloat16).contiguous(memory_format=cl)
+    w_perm = w.permute(1, 2, 3, 0).contiguous()
+    prev = os.environ.get("DDLB_CONV_V2")
+    try:
+        os.environ["DDLB_CONV_V2"] = "1"
+        y2 = ext.conv_igemm_fwd(x, w, stride, pad)
+        dx2 = ext.conv_igemm_dgrad(dy, w_perm, N, C, H, W, stride, pad)
+        os.environ["DDLB_CONV_V2"] = "0"
+        y1 = ext.conv_igemm_fwd(x, w, stride, pad)
+        dx1 = ext.conv_igemm_dgrad(dy, w_perm, N, C, H, W, stride, pad)
+    finally:
+        if prev is None:
+            os.environ.pop("DDLB_CONV_V2", None)
+        else:
+            os.environ["DDLB_CONV_V2"] = prev
+    assert torch.equal(y1, y2), "fwd structures disagree"
+    assert torch.equal(dx1, dx2), "dgrad structures disagree"
