@@ -183,28 +183,32 @@ DEV const bf16* b_slot_addr(const ConvParams& p, const Cursor& cu,
 }
 
 // ---- the kernel -------------------------------------------------------
-template <int MODE, int BN_, int WR, int WC>
+// COMB_B (BN_ == 64): the B slot is too small for one glds per thread
+// per k-half, so one stage call covers BOTH k-halves (waves 0-3 fill
+// k0, waves 4-7 fill k1) and B's cursor advances by 64.
+template <int MODE, int BM_, int BN_, int WR, int WC>
 __global__ __launch_bounds__(THREADS2, 2)
 void conv_igemm2_kernel(ConvParams p) {
-  constexpr int WM = BM2 / WR;       // wave tile M (128)
-  constexpr int WN = BN_ / WC;       // wave tile N (64 / 32)
-  constexpr int MF = WM / 16;        // acc M fragments (8)
-  constexpr int NF = WN / 16;        // acc N fragments (4 / 2)
-  constexpr int MH = MF / 2;         // M fragments per phase (4)
-  constexpr int GA = BM2 * 32 / 8 / THREADS2;  // A glds per wave/slot (2)
-  constexpr int GB = BN_ * 32 / 8 / THREADS2;  // B glds per wave/slot
-  constexpr int VN = GA + GB;        // counted vmcnt slack
+  constexpr int WM = BM_ / WR;       // wave tile M
+  constexpr int WN = BN_ / WC;       // wave tile N
+  constexpr int MF = WM / 16;        // acc M fragments
+  constexpr int NF = WN / 16;        // acc N fragments
+  constexpr int MH = MF / 2;         // M fragments per phase
+  constexpr int GA = BM_ * 32 / 8 / THREADS2;  // A glds per wave/slot
+  constexpr bool COMB_B = (BN_ * 32 / 8) < THREADS2;
+  constexpr int GB = COMB_B ? 1 : BN_ * 32 / 8 / THREADS2;
   static_assert(WR * WC == 8, "8 waves");
-  static_assert(GA >= 1 && GB >= 1, "tile too narrow for 512 threads");
+  static_assert(GA >= 1 && MH >= 1, "tile vs thread count");
+  static_assert(!COMB_B || BN_ == 64, "combined-B staging needs BN=64");
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* lds = reinterpret_cast<bf16*>(smem);
   // slot(buf, kh): A rows then B rows, 32 bf16 per row
   auto aslot = [&](int buf, int kh) {
-    return lds + ((buf * 2 + kh) * (BM2 + BN_)) * 32;
+    return lds + ((buf * 2 + kh) * (BM_ + BN_)) * 32;
   };
   auto bslot = [&](int buf, int kh) {
-    return aslot(buf, kh) + BM2 * 32;
+    return aslot(buf, kh) + BM_ * 32;
   };
 
   const int tid = threadIdx.x;
@@ -222,7 +226,7 @@ void conv_igemm2_kernel(ConvParams p) {
     block = (xcd < rmd ? xcd * (q + 1) : rmd * (q + 1) + (xcd - rmd) * q)
             + idx;
   }
-  const long bm = (long)(block / nbn) * BM2;
+  const long bm = (long)(block / nbn) * BM_;
   const long bn = (long)(block % nbn) * BN_;
 
   const int nsteps = (int)((p.Kd + BK2 - 1) / BK2);
@@ -230,11 +234,12 @@ void conv_igemm2_kernel(ConvParams p) {
   // ---- staging slots ---------------------------------------------------
   // chunk index within a slot: ca = (wid*G + j)*64 + lane
   //   row = ca>>2, stored pos = ca&3, source cg = (ca&3)^sigma((ca>>4)&3)
-  // Every slot of this thread shares the same cg, hence ONE cursor.
-  // Stage-call sequence is A(k),B(k),A(k+32),B(k+32),...: the cursor
-  // advances once per A/B pair, after the B call.
+  // Every slot of this thread shares the same cg. A and B each keep a
+  // cursor: A advances +32 per stage_a (k-half), B +32 per stage_b —
+  // or +64 when one combined call stages both halves.
   const int cg = (lane & 3) ^ sigma4((lane >> 4) & 3);
-  Cursor cur = cursor_init<MODE>(p, cg);
+  Cursor acur = cursor_init<MODE>(p, cg);
+  Cursor bcur = cursor_init<MODE>(p, cg + (COMB_B && wid >= 4 ? 4 : 0));
   unsigned abase[GA];
   int auv[GA];
   unsigned boff[GB];
@@ -246,26 +251,33 @@ void conv_igemm2_kernel(ConvParams p) {
   }
 #pragma unroll
   for (int j = 0; j < GB; ++j) {
-    const int cb = (wid * GB + j) * 64 + lane;
-    boff[j] = b_slot_init<MODE>(p, bn + (cb >> 2), cg, bok[j]);
+    const int cb = COMB_B ? (wid & 3) * 64 + lane
+                          : (wid * GB + j) * 64 + lane;
+    boff[j] = b_slot_init<MODE>(p, bn + (cb >> 2),
+                                cg + (COMB_B && wid >= 4 ? 4 : 0),
+                                bok[j]);
   }
 
   auto stage_a = [&](int buf, int kh) {
     bf16* la = aslot(buf, kh);
 #pragma unroll
     for (int j = 0; j < GA; ++j)
-      glds16(a_slot_addr<MODE>(p, cur, abase[j], auv[j]),
+      glds16(a_slot_addr<MODE>(p, acur, abase[j], auv[j]),
              la + ((wid * GA + j) * 64 + lane) * 8);
+    cursor_advance<MODE>(p, acur);
   };
+  // non-combined: stages one k-half. combined: waves 0-3 fill k0 and
+  // waves 4-7 fill k1 in ONE call (kh argument ignored).
   auto stage_b = [&](int buf, int kh) {
-    bf16* lb = bslot(buf, kh);
+    bf16* lb = bslot(buf, COMB_B ? (wid >= 4 ? 1 : 0) : kh);
 #pragma unroll
     for (int j = 0; j < GB; ++j) {
-      glds16(b_slot_addr<MODE>(p, cur, boff[j], bok[j]),
-             lb + ((wid * GB + j) * 64 + lane) * 8);
-      if (MODE != 2) boff[j] += 32;
+      glds16(b_slot_addr<MODE>(p, bcur, boff[j], bok[j]),
+             lb + (((COMB_B ? (wid & 3) : wid * GB + j)) * 64 + lane) * 8);
+      if (MODE != 2) boff[j] += COMB_B ? 64 : 32;
     }
-    cursor_advance<MODE>(p, cur);
+    cursor_advance<MODE>(p, bcur);
+    if (COMB_B) cursor_advance<MODE>(p, bcur);
   };
 
   // ---- fragment read geometry -----------------------------------------
@@ -287,10 +299,11 @@ void conv_igemm2_kernel(ConvParams p) {
 
   // ---- prologue: stage K-tile 0 fully ----------------------------------
   stage_a(0, 0);
-  stage_b(0, 0);
+  stage_b(0, 0);  // combined form also covers k1 here
   stage_a(0, 1);
-  stage_b(0, 1);
-  wait_vmcnt<VN>();  // A(k0),B(k0) landed; k1 halves may be in flight
+  if (!COMB_B) stage_b(0, 1);
+  // A(k0),B(k0) landed; later halves may stay in flight
+  wait_vmcnt<COMB_B ? GA : GA + GB>();
   __builtin_amdgcn_s_barrier();
 
   // ---- main loop: 4 phases per K-tile ----------------------------------
@@ -326,15 +339,24 @@ void conv_igemm2_kernel(ConvParams p) {
     __builtin_amdgcn_s_barrier();                                         \
   } while (0)
 
+  // counted waits: end-of-ph1 guards this tile's k1 reads (allows the
+  // two units issued at ph0/ph1 to stay in flight); end-of-ph3 guards
+  // the next tile's k0 reads (allows ph2/ph3's units)
+  constexpr int V1 = COMB_B ? GA + 1 : GA + GB;
+  constexpr int V3 = COMB_B ? GA : GA + GB;
   for (int t = 0; t < nsteps; ++t) {
     const int cbuf = t & 1;
     const int nxt = cbuf ^ 1;
     const bool more = (t + 1 < nsteps);
     if (more) {
       PH(0, 0, stage_a(nxt, 0), );
-      PH(1, 0, stage_b(nxt, 0), wait_vmcnt<VN>());
+      PH(1, 0, stage_b(nxt, 0), wait_vmcnt<V1>());
       PH(0, 1, stage_a(nxt, 1), );
-      PH(1, 1, stage_b(nxt, 1), wait_vmcnt<VN>());
+      if (COMB_B) {
+        PH(1, 1, , wait_vmcnt<V3>());
+      } else {
+        PH(1, 1, stage_b(nxt, 1), wait_vmcnt<V3>());
+      }
     } else {
       // last K-tile: nothing left to stage; the k1 halves may still be
       // in flight, so the mid-tile wait drains fully
@@ -346,12 +368,37 @@ void conv_igemm2_kernel(ConvParams p) {
   }
 #undef PH
 
-  // ---- epilogue: D[row=fk*4+reg][col=fr] per fragment -------------------
+  // ---- epilogue: LDS transpose -> full-width coalesced stores ----------
+  // The MFMA D layout (row = fk*4+reg per lane, col = fr across lanes)
+  // would store 2 B/lane in 32-B row segments (~2x store-bandwidth
+  // waste — these big-M/small-Kd convs are store-bound). Round-trip the
+  // tile through the now-free LDS (row-major [BM_][BN_] bf16 image; its
+  // 256-B rows make the b128 read-back groups conflict-free) and emit
+  // 16-B/lane stores covering full rows.
+  wait_vmcnt<0>();
+  __builtin_amdgcn_s_barrier();
+  static_assert((long)BM_ * BN_ * 2 <= 4L * (BM_ + BN_) * 32 * 2,
+                "tile image must fit the staging LDS");
+  {
+    bf16* img = lds;
 #pragma unroll
-  for (int mf = 0; mf < MF; ++mf) {
+    for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const long row = bm + awoff + mf * 16 + fk * 4 + reg;
+      for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+          img[(awoff + mf * 16 + fk * 4 + reg) * BN_ +
+              bwoff + nf * 16 + fr] = from_f32<bf16>(acc[mf][nf][reg]);
+    __syncthreads();
+    constexpr int CPR = BN_ / 8;            // 16-B chunks per tile row
+    constexpr int ITERS = BM_ * CPR / THREADS2;
+    typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      const int idx = it * THREADS2 + tid;
+      const int r = idx / CPR;
+      const int ck = idx - r * CPR;
+      const long row = bm + r;
       if (row >= p.M) continue;
       long out_row = row;
       if (MODE == 2) {
@@ -363,11 +410,14 @@ void conv_igemm2_kernel(ConvParams p) {
         out_row = ((long)n * p.H + 2 * ii + p.cls_a) * p.W
                   + 2 * jj + p.cls_b;
       }
-#pragma unroll
-      for (int nf = 0; nf < NF; ++nf) {
-        const long col = bn + bwoff + nf * 16 + fr;
-        if (col < p.Nd)
-          p.out[out_row * p.Nd + col] = from_f32<bf16>(acc[mf][nf][reg]);
+      const long colbase = bn + ck * 8;
+      const bf16* src = img + r * BN_ + ck * 8;
+      if (colbase + 8 <= p.Nd) {
+        *reinterpret_cast<u16x8*>(p.out + out_row * p.Nd + colbase) =
+            *reinterpret_cast<const u16x8*>(src);
+      } else if (colbase < p.Nd) {
+        for (int j = 0; j < (int)(p.Nd - colbase); ++j)
+          p.out[out_row * p.Nd + colbase + j] = src[j];
       }
     }
   }
@@ -375,21 +425,23 @@ void conv_igemm2_kernel(ConvParams p) {
 
 template <int MODE>
 bool dispatch_v2(const ConvParams& p, hipStream_t stream) {
-  const long nbm = (p.M + BM2 - 1) / BM2;
-#define LAUNCH2(BN_, WR, WC)                                                \
+#define LAUNCH2(BM_, BN_, WR, WC)                                           \
   do {                                                                      \
+    const long nbm = (p.M + (BM_) - 1) / (BM_);                             \
     const long nbn = (p.Nd + (BN_) - 1) / (BN_);                            \
-    const size_t lds_bytes = 4 * (BM2 + (BN_)) * 32 * sizeof(bf16);         \
-    hipLaunchKernelGGL((conv_igemm2_kernel<MODE, BN_, WR, WC>),             \
+    const size_t lds_bytes = 4 * ((BM_) + (BN_)) * 32 * sizeof(bf16);       \
+    hipLaunchKernelGGL((conv_igemm2_kernel<MODE, BM_, BN_, WR, WC>),        \
                        dim3((unsigned)(nbm * nbn)), dim3(THREADS2),         \
                        lds_bytes, stream, p);                               \
   } while (0)
   // BN=256 (acc 128 regs/wave) cannot fit beside the im2col staging
   // state in the 256-VGPR/2-wave budget (measured 105-reg spill); the
-  // 256x128 tile (acc 64, 204 VGPRs clean) serves all Nd >= 96 with
-  // column blocks.
+  // 256x128 tile (acc 64, ~204 VGPRs clean) serves all Nd >= 96 with
+  // column blocks. Nd in [48,96) gets a 512x64 tile (combined-B).
   if (p.Nd >= 96) {
-    LAUNCH2(128, 2, 4);
+    LAUNCH2(256, 128, 2, 4);
+  } else if (p.Nd >= 48) {
+    LAUNCH2(512, 64, 8, 1);
   } else {
     return false;
   }

@@ -410,6 +410,8 @@ def test_model_trains_on_native_kernels():
     (3, 96, 13, 13, 104, 3, 1, 1),
     (2, 128, 9, 9, 512, 1, 1, 0),
     (2, 256, 10, 10, 128, 3, 2, 1),
+    (3, 64, 19, 19, 64, 3, 1, 1),    # narrow (512x64 combined-B) variant
+    (2, 64, 10, 10, 56, 1, 1, 0),    # narrow, Nd=56 tail
 ])
 def test_conv_v2_matches_v1(shape):
     """The deep-pipeline (v2) and 128-tile (v1) structures accumulate
