@@ -28,19 +28,21 @@ void set_bn_variant(int v);
 void launch_bn_finalize(double*, float*, float*, float*, float*,
                         int64_t, double, float, float, hipStream_t);
 template <typename T>
-void launch_bn_apply(const T*, const T*, T*, const float*, const float*,
-                     const float*, const float*, int64_t, int64_t, int64_t,
-                     int, int, hipStream_t);
+bool launch_bn_apply(const T*, const T*, T*, const float*, const float*,
+                     const float*, const float*, unsigned char*, int64_t,
+                     int64_t, int64_t, int, int, hipStream_t);
 template <typename T>
 void launch_bn_bwd_reduce(const T*, const T*, const T*, const float*,
-                          const float*, double*, int64_t, int64_t, int64_t,
+                          const float*, double*, const unsigned char*,
+                          int64_t, int64_t, int64_t,
                           int, int, hipStream_t);
 void launch_bn_bwd_finalize(double*, const float*, const float*,
                             float*, float*, float*, int64_t, double, int,
                             hipStream_t);
 template <typename T>
 void launch_bn_bwd_dx(const T*, const T*, const T*, const float*,
-                      const float*, const float*, T*, T*, int64_t, int64_t,
+                      const float*, const float*, T*, T*,
+                      const unsigned char*, int64_t, int64_t,
                       int64_t, int, int, hipStream_t);
 
 template <typename T>
@@ -193,28 +195,43 @@ std::vector<torch::Tensor> bn_act_fwd(
   }
   torch::Tensor y = torch::empty_like(x);
   const int64_t total = x.numel();
+  // 1-bit activation mask for the backward (bf16 NHWC training path):
+  // bit (i & 7) of byte i/8 gates element i's activation gradient.
+  torch::Tensor msk;
+  const bool want_mask =
+      training && act != 0 && nhwc && is_bf16(x) && C % 8 == 0;
+  if (want_mask)
+    msk = torch::empty({total / 8}, x.options().dtype(at::kByte));
+  bool mask_written = false;
   if (is_bf16(x))
-    launch_bn_apply<__hip_bfloat16>(
+    mask_written = launch_bn_apply<__hip_bfloat16>(
         dptr<__hip_bfloat16>(x),
         res ? dptr<__hip_bfloat16>(*res) : nullptr, dptr<__hip_bfloat16>(y),
         dptr<float>(mean), dptr<float>(invstd), dptr<float>(gamma),
-        dptr<float>(beta), C, HW, total, (int)act, nhwc, s);
+        dptr<float>(beta),
+        want_mask ? msk.data_ptr<unsigned char>() : nullptr,
+        C, HW, total, (int)act, nhwc, s);
   else
     launch_bn_apply<float>(dptr<float>(x),
                            res ? dptr<float>(*res) : nullptr, dptr<float>(y),
                            dptr<float>(mean), dptr<float>(invstd),
-                           dptr<float>(gamma), dptr<float>(beta), C, HW,
-                           total, (int)act, nhwc, s);
+                           dptr<float>(gamma), dptr<float>(beta), nullptr,
+                           C, HW, total, (int)act, nhwc, s);
+  if (mask_written) return {y, mean, invstd, msk};
   return {y, mean, invstd};
 }
 
-std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy,
+                                      c10::optional<torch::Tensor> y_opt,
                                       torch::Tensor x, torch::Tensor mean,
                                       torch::Tensor invstd,
                                       torch::Tensor gamma, int64_t act,
                                       bool training, bool need_dres,
-                                      bool nhwc) {
+                                      bool nhwc,
+                                      c10::optional<torch::Tensor> msk) {
   TORCH_CHECK(dy.is_cuda(), "dy must be on the HIP device");
+  TORCH_CHECK(y_opt || msk,
+              "bn_act_bwd needs y or the 1-bit activation mask");
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto s = cur_stream();
   auto fopt = x.options().dtype(at::kFloat);
@@ -226,34 +243,42 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy, torch::Tensor y,
   torch::Tensor dres =
       need_dres ? torch::empty_like(x) : torch::Tensor();
   const int64_t total = x.numel();
+  const unsigned char* mp =
+      msk ? msk->data_ptr<unsigned char>() : nullptr;
   if (is_bf16(x)) {
+    TORCH_CHECK(mp || y_opt, "masked path needs the mask tensor");
+    const __hip_bfloat16* yp =
+        y_opt ? dptr<__hip_bfloat16>(*y_opt) : nullptr;
     launch_bn_bwd_reduce<__hip_bfloat16>(
-        dptr<__hip_bfloat16>(dy), dptr<__hip_bfloat16>(y),
+        dptr<__hip_bfloat16>(dy), yp,
         dptr<__hip_bfloat16>(x), dptr<float>(mean), dptr<float>(invstd),
-        dptr<double>(sums), N, C, HW, (int)act, nhwc, s);
+        dptr<double>(sums), mp, N, C, HW, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
                            dptr<float>(dbeta), dptr<float>(k), C,
                            (double)(N * HW), training, s);
     launch_bn_bwd_dx<__hip_bfloat16>(
-        dptr<__hip_bfloat16>(dy), dptr<__hip_bfloat16>(y),
+        dptr<__hip_bfloat16>(dy), yp,
         dptr<__hip_bfloat16>(x), dptr<float>(mean), dptr<float>(invstd),
         dptr<float>(k), dptr<__hip_bfloat16>(dx),
-        need_dres ? dptr<__hip_bfloat16>(dres) : nullptr, C, HW, total,
+        need_dres ? dptr<__hip_bfloat16>(dres) : nullptr, mp, C, HW, total,
         (int)act, nhwc, s);
   } else {
-    launch_bn_bwd_reduce<float>(dptr<float>(dy), dptr<float>(y),
+    TORCH_CHECK(y_opt, "fp32 backward needs y");
+    launch_bn_bwd_reduce<float>(dptr<float>(dy), dptr<float>(*y_opt),
                                 dptr<float>(x), dptr<float>(mean),
-                                dptr<float>(invstd), dptr<double>(sums), N,
-                                C, HW, (int)act, nhwc, s);
+                                dptr<float>(invstd), dptr<double>(sums),
+                                nullptr, N, C, HW, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
                            dptr<float>(dbeta), dptr<float>(k), C,
                            (double)(N * HW), training, s);
-    launch_bn_bwd_dx<float>(dptr<float>(dy), dptr<float>(y), dptr<float>(x),
+    launch_bn_bwd_dx<float>(dptr<float>(dy), dptr<float>(*y_opt),
+                            dptr<float>(x),
                             dptr<float>(mean), dptr<float>(invstd),
                             dptr<float>(k), dptr<float>(dx),
-                            need_dres ? dptr<float>(dres) : nullptr, C, HW,
+                            need_dres ? dptr<float>(dres) : nullptr,
+                            nullptr, C, HW,
                             total, (int)act, nhwc, s);
   }
   if (need_dres) return {dx, dgamma, dbeta, dres};

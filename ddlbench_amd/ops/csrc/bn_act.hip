@@ -396,6 +396,7 @@ __global__ void bn_apply_nhwc_kernel2(const T* __restrict__ x,
                                       const float* __restrict__ invstd,
                                       const float* __restrict__ gamma,
                                       const float* __restrict__ beta,
+                                      unsigned char* __restrict__ msk,
                                       int64_t rows, int64_t C, int CG8) {
   typedef __attribute__((ext_vector_type(8))) short short8x;
   const int ci = threadIdx.x % CG8;
@@ -421,6 +422,7 @@ __global__ void bn_apply_nhwc_kernel2(const T* __restrict__ x,
     short8x vr;
     if (ADD) vr = rv[i8];
     short8x vy;
+    unsigned mbits = 0;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       __hip_bfloat16 h;
@@ -432,15 +434,21 @@ __global__ void bn_apply_nhwc_kernel2(const T* __restrict__ x,
         __builtin_memcpy(&h, &u, 2);
         v += __bfloat162float(h);
       }
+      if (ACT != 0 && act_mask<ACT>(v) > 0.f) mbits |= 1u << j;
       h = __float2bfloat16(act_fwd<ACT>(v));
       __builtin_memcpy(&u, &h, 2);
       vy[j] = (short)u;
     }
     yv[i8] = vy;
+    // 1-bit activation mask: backward re-reads this byte instead of y
+    // (docs/ROADMAP.md item 7: ~2 of 7 memory passes saved)
+    if (ACT != 0 && msk != nullptr) msk[i8] = (unsigned char)mbits;
   }
 }
 
-template <typename T, int ACT, bool ADD>
+// MASKED: the activation gradient gate comes from the 1-bit mask the
+// forward apply wrote (one byte per 8 channels); y is never read.
+template <typename T, int ACT, bool ADD, bool MASKED = false>
 __global__ void bn_bwd_dx_nhwc_kernel2(const T* __restrict__ dy,
                                        const T* __restrict__ y,
                                        const T* __restrict__ x,
@@ -448,7 +456,9 @@ __global__ void bn_bwd_dx_nhwc_kernel2(const T* __restrict__ dy,
                                        const float* __restrict__ invstd,
                                        const float* __restrict__ k,
                                        T* __restrict__ dx,
-                                       T* __restrict__ dres, int64_t rows,
+                                       T* __restrict__ dres,
+                                       const unsigned char* __restrict__ msk,
+                                       int64_t rows,
                                        int64_t C, int CG8) {
   typedef __attribute__((ext_vector_type(8))) short short8x;
   const int ci = threadIdx.x % CG8;
@@ -475,7 +485,11 @@ __global__ void bn_bwd_dx_nhwc_kernel2(const T* __restrict__ dy,
   short8x* drv = reinterpret_cast<short8x*>(dres);
   for (int64_t r = begin + rj; r < end; r += RG) {
     const int64_t i8 = (r * C + c0) / 8;
-    short8x vdy = dyv[i8], vy = yv[i8], vx = xv[i8];
+    short8x vdy = dyv[i8], vx = xv[i8];
+    short8x vy;
+    unsigned mb = 0;
+    if (MASKED) mb = msk[i8];
+    else vy = yv[i8];
     short8x vdx, vdr;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -484,13 +498,18 @@ __global__ void bn_bwd_dx_nhwc_kernel2(const T* __restrict__ dy,
       u = (unsigned short)vdy[j];
       __builtin_memcpy(&h, &u, 2);
       const float fdy = __bfloat162float(h);
-      u = (unsigned short)vy[j];
-      __builtin_memcpy(&h, &u, 2);
-      const float fy = __bfloat162float(h);
+      float gate;
+      if (MASKED) {
+        gate = (float)((mb >> j) & 1u);
+      } else {
+        u = (unsigned short)vy[j];
+        __builtin_memcpy(&h, &u, 2);
+        gate = act_mask<ACT>(__bfloat162float(h));
+      }
       u = (unsigned short)vx[j];
       __builtin_memcpy(&h, &u, 2);
       const float fx = __bfloat162float(h);
-      const float g = fdy * act_mask<ACT>(fy);
+      const float g = fdy * gate;
       const float xhat = (fx - mu[j]) * is[j];
       h = __float2bfloat16(k1[j] * (g - k2[j] - xhat * k3[j]));
       __builtin_memcpy(&u, &h, 2);
@@ -542,13 +561,15 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
 }
 
 // ---- backward reduce, NHWC (same block geometry as the stats) ---------
-template <typename T, int ACT>
+template <typename T, int ACT, bool MASKED = false>
 __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
                                           const T* __restrict__ y,
                                           const T* __restrict__ x,
                                           const float* __restrict__ mean,
                                           const float* __restrict__ invstd,
                                           double* __restrict__ sums,
+                                          const unsigned char* __restrict__
+                                              msk,
                                           int64_t rows, int64_t C, int CG) {
   __shared__ double tmp[2 * 256];
   const int ci = threadIdx.x % CG;
@@ -562,9 +583,15 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
   double sdy = 0.0, sdyx = 0.0;
   if (active) {
     const float mu = mean[c], is = invstd[c];
+    const int cbit = (int)(c & 7);
     for (int64_t r = begin + rj; r < end; r += RG) {
       const int64_t idx = r * C + c;
-      const float g = to_f32(dy[idx]) * act_mask<ACT>(to_f32(y[idx]));
+      float gate;
+      if (MASKED)
+        gate = (float)((msk[idx >> 3] >> cbit) & 1u);
+      else
+        gate = act_mask<ACT>(to_f32(y[idx]));
+      const float g = to_f32(dy[idx]) * gate;
       const float xhat = (to_f32(x[idx]) - mu) * is;
       sdy += g;
       sdyx += fma((double)g, (double)xhat, 0.0);
@@ -583,11 +610,12 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
   }
 }
 
-template <typename T, int ACT>
+template <typename T, int ACT, bool MASKED = false>
 __global__ void bn_bwd_reduce_nhwc_vec_kernel(
     const T* __restrict__ dy, const T* __restrict__ y,
     const T* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, double* __restrict__ sums,
+    const unsigned char* __restrict__ msk,
     int64_t rows, int64_t C, int CG8) {
   typedef __attribute__((ext_vector_type(8))) short short8x;
   __shared__ float tmp[2][256][8];
@@ -613,7 +641,11 @@ __global__ void bn_bwd_reduce_nhwc_vec_kernel(
     const short8x* xv = reinterpret_cast<const short8x*>(x);
     for (int64_t r = begin + rj; r < end; r += RG) {
       const int64_t i8 = (r * C + c0) / 8;
-      short8x vdy = dyv[i8], vy = yv[i8], vx = xv[i8];
+      short8x vdy = dyv[i8], vx = xv[i8];
+      short8x vy;
+      unsigned mb = 0;
+      if (MASKED) mb = msk[i8];
+      else vy = yv[i8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         __hip_bfloat16 h;
@@ -621,13 +653,18 @@ __global__ void bn_bwd_reduce_nhwc_vec_kernel(
         u = (unsigned short)vdy[j];
         __builtin_memcpy(&h, &u, 2);
         const float fdy = __bfloat162float(h);
-        u = (unsigned short)vy[j];
-        __builtin_memcpy(&h, &u, 2);
-        const float fy = __bfloat162float(h);
+        float gate;
+        if (MASKED) {
+          gate = (float)((mb >> j) & 1u);
+        } else {
+          u = (unsigned short)vy[j];
+          __builtin_memcpy(&h, &u, 2);
+          gate = act_mask<ACT>(__bfloat162float(h));
+        }
         u = (unsigned short)vx[j];
         __builtin_memcpy(&h, &u, 2);
         const float fx = __bfloat162float(h);
-        const float g = fdy * act_mask<ACT>(fy);
+        const float g = fdy * gate;
         sdy[j] += g;
         sdyx[j] = fmaf(g, (fx - mu[j]) * is[j], sdyx[j]);
       }
@@ -750,9 +787,10 @@ void launch_bn_finalize(double* sums, float* mean, float* invstd,
 }
 
 template <typename T>
-void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
+bool launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
                      const float* invstd, const float* gamma,
-                     const float* beta, int64_t C, int64_t HW, int64_t total,
+                     const float* beta, unsigned char* msk, int64_t C,
+                     int64_t HW, int64_t total,
                      int act, int nhwc, hipStream_t stream) {
   const int block = 256;
   const int64_t cdiv = nhwc ? 1 : HW;
@@ -767,13 +805,13 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
 #define KCASE(ACT, ADD)                                                     \
     hipLaunchKernelGGL((bn_apply_nhwc_kernel2<T, ACT, ADD>),                \
                        dim3(cblocks, S), dim3(block), 0, stream, x, res,    \
-                       y, mean, invstd, gamma, beta, rows, C, CG8)
+                       y, mean, invstd, gamma, beta, msk, rows, C, CG8)
     if (act == 0) { if (add) KCASE(0, true); else KCASE(0, false); }
     else if (act == 1) { if (add) KCASE(1, true); else KCASE(1, false); }
     else { if (add) KCASE(2, true); else KCASE(2, false); }
 #undef KCASE
     HIP_CHECK_LAST();
-    return;
+    return act != 0 && msk != nullptr;  // mask written on this path
   }
   if (vec8ok) {
     const int64_t total8 = total / 8;
@@ -794,7 +832,7 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
 #undef VSEL
 #undef VCASE
     HIP_CHECK_LAST();
-    return;
+    return false;
   }
   const int grid = elementwise_grid(total, block);
 #define CASE(ACT, ADD)                                                       \
@@ -810,41 +848,44 @@ void launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
   else CASE(2, true);
 #undef CASE
   HIP_CHECK_LAST();
+  return false;
 }
 
 template <typename T>
 void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
                           const float* mean, const float* invstd,
-                          double* sums, int64_t N, int64_t C, int64_t HW,
+                          double* sums, const unsigned char* msk,
+                          int64_t N, int64_t C, int64_t HW,
                           int act, int nhwc, hipStream_t stream) {
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
+    const bool masked = msk != nullptr && act != 0;
     if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
       int64_t S = i64min(i64max(rows / 512, 1),
                          i64max(2048 / cblocks, 1));
-#define CASE(ACT)                                                           \
-      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_vec_kernel<T, ACT>),           \
+#define CASE(ACT, MSK)                                                      \
+      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_vec_kernel<T, ACT, MSK>),      \
                          dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
-                         x, mean, invstd, sums, rows, C, CG8)
-      if (act == 0) CASE(0);
-      else if (act == 1) CASE(1);
-      else CASE(2);
+                         x, mean, invstd, sums, msk, rows, C, CG8)
+      if (act == 0) CASE(0, false);
+      else if (act == 1) { if (masked) CASE(1, true); else CASE(1, false); }
+      else { if (masked) CASE(2, true); else CASE(2, false); }
 #undef CASE
     } else {
       const int CG = C >= 64 ? 64 : (int)C;
       const int64_t cblocks = (C + CG - 1) / CG;
       int64_t S = i64min(i64max(rows / 512, 1),
                          i64max(2048 / cblocks, 1));
-#define CASE(ACT)                                                           \
-      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT>),               \
+#define CASE(ACT, MSK)                                                      \
+      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT, MSK>),          \
                          dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
-                         x, mean, invstd, sums, rows, C, CG)
-      if (act == 0) CASE(0);
-      else if (act == 1) CASE(1);
-      else CASE(2);
+                         x, mean, invstd, sums, msk, rows, C, CG)
+      if (act == 0) CASE(0, false);
+      else if (act == 1) { if (masked) CASE(1, true); else CASE(1, false); }
+      else { if (masked) CASE(2, true); else CASE(2, false); }
 #undef CASE
     }
   } else {
@@ -876,6 +917,7 @@ void launch_bn_bwd_finalize(double* sums, const float* gamma,
 template <typename T>
 void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
                       const float* invstd, const float* k, T* dx, T* dres,
+                      const unsigned char* msk,
                       int64_t C, int64_t HW, int64_t total, int act,
                       int nhwc, hipStream_t stream) {
   const int block = 256;
@@ -888,13 +930,20 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
     int64_t S = i64min(i64max(rows / 256, 1), i64max(2048 / cblocks, 1));
     const bool add = dres != nullptr;
-#define KCASE(ACT, ADD)                                                     \
-    hipLaunchKernelGGL((bn_bwd_dx_nhwc_kernel2<T, ACT, ADD>),               \
+    const bool masked = msk != nullptr && act != 0;
+#define KCASE(ACT, ADD, MSK)                                                \
+    hipLaunchKernelGGL((bn_bwd_dx_nhwc_kernel2<T, ACT, ADD, MSK>),          \
                        dim3(cblocks, S), dim3(block), 0, stream, dy, y, x,  \
-                       mean, invstd, k, dx, dres, rows, C, CG8)
-    if (act == 0) { if (add) KCASE(0, true); else KCASE(0, false); }
-    else if (act == 1) { if (add) KCASE(1, true); else KCASE(1, false); }
-    else { if (add) KCASE(2, true); else KCASE(2, false); }
+                       mean, invstd, k, dx, dres, msk, rows, C, CG8)
+#define KSEL(ACT)                                                           \
+    do { if (add) { if (masked) KCASE(ACT, true, true);                     \
+                    else KCASE(ACT, true, false); }                         \
+         else { if (masked) KCASE(ACT, false, true);                        \
+                else KCASE(ACT, false, false); } } while (0)
+    if (act == 0) KSEL(0);
+    else if (act == 1) KSEL(1);
+    else KSEL(2);
+#undef KSEL
 #undef KCASE
     HIP_CHECK_LAST();
     return;
@@ -939,17 +988,20 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
 #define INSTANTIATE(T)                                                        \
   template void launch_bn_stats<T>(const T*, double*, int64_t, int64_t,       \
                                    int64_t, int, hipStream_t);                \
-  template void launch_bn_apply<T>(const T*, const T*, T*, const float*,      \
+  template bool launch_bn_apply<T>(const T*, const T*, T*, const float*,      \
                                    const float*, const float*, const float*,  \
+                                   unsigned char*,                            \
                                    int64_t, int64_t, int64_t, int, int,       \
                                    hipStream_t);                              \
   template void launch_bn_bwd_reduce<T>(const T*, const T*, const T*,         \
                                         const float*, const float*, double*,  \
+                                        const unsigned char*,                 \
                                         int64_t, int64_t, int64_t, int, int,  \
                                         hipStream_t);                         \
   template void launch_bn_bwd_dx<T>(const T*, const T*, const T*,             \
                                     const float*, const float*, const float*, \
-                                    T*, T*, int64_t, int64_t, int64_t, int,   \
+                                    T*, T*, const unsigned char*,             \
+                                    int64_t, int64_t, int64_t, int,           \
                                     int, hipStream_t);
 
 INSTANTIATE(float)

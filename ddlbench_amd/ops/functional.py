@@ -45,10 +45,18 @@ class _FusedBNAct(torch.autograd.Function):
         nhwc = _is_nhwc(x)
         x = _layout_contiguous(x, nhwc)
         res = _layout_contiguous(res, nhwc) if res is not None else None
-        y, mean, invstd = ext.bn_act_fwd(x, res, gamma, beta, running_mean,
-                                         running_var, training, momentum,
-                                         eps, act_code, nhwc)
-        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        outs = ext.bn_act_fwd(x, res, gamma, beta, running_mean,
+                              running_var, training, momentum,
+                              eps, act_code, nhwc)
+        y, mean, invstd = outs[0], outs[1], outs[2]
+        if len(outs) > 3:
+            # 1-bit activation mask: backward never reads y (saves two
+            # full-tensor passes AND the y activation memory)
+            ctx.save_for_backward(x, mean, invstd, gamma, outs[3])
+            ctx.masked = True
+        else:
+            ctx.save_for_backward(x, mean, invstd, gamma, y)
+            ctx.masked = False
         ctx.training = training
         ctx.act_code = act_code
         ctx.has_res = res is not None
@@ -58,10 +66,11 @@ class _FusedBNAct(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = _ops.require_extension()
-        x, y, mean, invstd, gamma = ctx.saved_tensors
+        x, mean, invstd, gamma, aux = ctx.saved_tensors
+        y, mask = (None, aux) if ctx.masked else (aux, None)
         out = ext.bn_act_bwd(_layout_contiguous(dy, ctx.nhwc), y, x, mean,
                              invstd, gamma, ctx.act_code, ctx.training,
-                             ctx.has_res, ctx.nhwc)
+                             ctx.has_res, ctx.nhwc, mask)
         dx, dgamma, dbeta = out[0], out[1], out[2]
         dres = out[3] if ctx.has_res else None
         return (dx, dgamma, dbeta, dres, None, None, None, None, None, None)
