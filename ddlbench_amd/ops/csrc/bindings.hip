@@ -22,11 +22,14 @@ void launch_fused_adam(uintptr_t*, uintptr_t*, uintptr_t*, uintptr_t*,
                        float, float, float, float, int, hipStream_t);
 
 template <typename T>
-void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int,
-                     hipStream_t);
+void launch_bn_stats(const T*, double*, int64_t, int64_t, int64_t, int64_t,
+                     int, hipStream_t);
 void set_bn_variant(int v);
+int64_t bn_reduce_gridS(int64_t N, int64_t C, int64_t HW, int nhwc,
+                        int elsize);
 void launch_bn_finalize(double*, float*, float*, float*, float*,
-                        int64_t, double, float, float, hipStream_t);
+                        int64_t, int64_t, double, float, float,
+                        hipStream_t);
 template <typename T>
 bool launch_bn_apply(const T*, const T*, T*, const float*, const float*,
                      const float*, const float*, unsigned char*, int64_t,
@@ -34,11 +37,11 @@ bool launch_bn_apply(const T*, const T*, T*, const float*, const float*,
 template <typename T>
 void launch_bn_bwd_reduce(const T*, const T*, const T*, const float*,
                           const float*, double*, const unsigned char*,
-                          int64_t, int64_t, int64_t,
+                          int64_t, int64_t, int64_t, int64_t,
                           int, int, hipStream_t);
 void launch_bn_bwd_finalize(double*, const float*, const float*,
-                            float*, float*, float*, int64_t, double, int,
-                            hipStream_t);
+                            float*, float*, float*, int64_t, int64_t,
+                            double, int, hipStream_t);
 template <typename T>
 void launch_bn_bwd_dx(const T*, const T*, const T*, const float*,
                       const float*, const float*, T*, T*,
@@ -97,16 +100,20 @@ T* dptr(const torch::Tensor& t) {
   return reinterpret_cast<T*>(t.data_ptr());
 }
 
-// Per-device BN reduction workspace. The finalize kernels write zeros
-// back after consuming, so after the first allocation no per-call fill
-// kernel runs (stream ordering serializes stats -> finalize -> reuse).
-torch::Tensor bn_sums_workspace(const torch::Tensor& like, int64_t C) {
+// Per-device BN reduction workspace: [S][2][C] double partial slabs.
+// Reduce-style kernels write plain per-block stores and the finalize
+// kernels sum over S — no cross-block atomics (a C=64 layer at S~1500
+// slices serialized ~1500 f64 atomics per channel address), and no
+// zeroing (every slab slot is written every call).
+torch::Tensor bn_sums_workspace(const torch::Tensor& like, int64_t C,
+                                int64_t S) {
   static std::unordered_map<int, torch::Tensor> ws;
   const int dev = (int)like.get_device();
+  const int64_t need = 2 * C * S;
   auto it = ws.find(dev);
-  if (it == ws.end() || it->second.numel() < 2 * C) {
-    const int64_t cap = std::max<int64_t>(C, 2048);
-    ws[dev] = torch::zeros({2 * cap}, like.options().dtype(at::kDouble));
+  if (it == ws.end() || it->second.numel() < need) {
+    const int64_t cap = std::max<int64_t>(need, 1 << 18);
+    ws[dev] = torch::empty({cap}, like.options().dtype(at::kDouble));
     it = ws.find(dev);
   }
   return it->second;
@@ -175,17 +182,20 @@ std::vector<torch::Tensor> bn_act_fwd(
   torch::Tensor mean = torch::empty({C}, fopt);
   torch::Tensor invstd = torch::empty({C}, fopt);
   if (training) {
-    torch::Tensor sums = bn_sums_workspace(x, C);
+    const int64_t S =
+        bn_reduce_gridS(N, C, HW, nhwc, (int)x.element_size());
+    torch::Tensor sums = bn_sums_workspace(x, C, S);
     if (is_bf16(x))
       launch_bn_stats<__hip_bfloat16>(dptr<__hip_bfloat16>(x),
-                                      dptr<double>(sums), N, C, HW, nhwc, s);
+                                      dptr<double>(sums), N, C, HW, S,
+                                      nhwc, s);
     else
       launch_bn_stats<float>(dptr<float>(x), dptr<double>(sums), N, C, HW,
-                             nhwc, s);
+                             S, nhwc, s);
     launch_bn_finalize(
         dptr<double>(sums), dptr<float>(mean), dptr<float>(invstd),
         running_mean ? dptr<float>(*running_mean) : nullptr,
-        running_var ? dptr<float>(*running_var) : nullptr, C,
+        running_var ? dptr<float>(*running_var) : nullptr, C, S,
         (double)(N * HW), (float)eps, (float)momentum, s);
   } else {
     TORCH_CHECK(running_mean && running_var,
@@ -235,7 +245,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy,
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto s = cur_stream();
   auto fopt = x.options().dtype(at::kFloat);
-  torch::Tensor sums = bn_sums_workspace(x, C);
+  const int64_t S = bn_reduce_gridS(N, C, HW, nhwc, (int)x.element_size());
+  torch::Tensor sums = bn_sums_workspace(x, C, S);
   torch::Tensor dgamma = torch::empty({C}, fopt);
   torch::Tensor dbeta = torch::empty({C}, fopt);
   torch::Tensor k = torch::empty({3, C}, fopt);
@@ -252,10 +263,10 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy,
     launch_bn_bwd_reduce<__hip_bfloat16>(
         dptr<__hip_bfloat16>(dy), yp,
         dptr<__hip_bfloat16>(x), dptr<float>(mean), dptr<float>(invstd),
-        dptr<double>(sums), mp, N, C, HW, (int)act, nhwc, s);
+        dptr<double>(sums), mp, N, C, HW, S, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
-                           dptr<float>(dbeta), dptr<float>(k), C,
+                           dptr<float>(dbeta), dptr<float>(k), C, S,
                            (double)(N * HW), training, s);
     launch_bn_bwd_dx<__hip_bfloat16>(
         dptr<__hip_bfloat16>(dy), yp,
@@ -268,10 +279,10 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor dy,
     launch_bn_bwd_reduce<float>(dptr<float>(dy), dptr<float>(*y_opt),
                                 dptr<float>(x), dptr<float>(mean),
                                 dptr<float>(invstd), dptr<double>(sums),
-                                nullptr, N, C, HW, (int)act, nhwc, s);
+                                nullptr, N, C, HW, S, (int)act, nhwc, s);
     launch_bn_bwd_finalize(dptr<double>(sums), dptr<float>(gamma),
                            dptr<float>(invstd), dptr<float>(dgamma),
-                           dptr<float>(dbeta), dptr<float>(k), C,
+                           dptr<float>(dbeta), dptr<float>(k), C, S,
                            (double)(N * HW), training, s);
     launch_bn_bwd_dx<float>(dptr<float>(dy), dptr<float>(*y_opt),
                             dptr<float>(x),

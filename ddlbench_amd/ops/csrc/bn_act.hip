@@ -64,8 +64,8 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, double* __restrict__ su
   __syncthreads();
   ss = block_reduce(ss, tmp, op, 0.0);
   if (threadIdx.x == 0) {
-    atomicAdd(&sums[c], s);
-    atomicAdd(&sums[C + c], ss);
+    sums[((int64_t)blockIdx.y * 2 + 0) * C + c] = s;
+    sums[((int64_t)blockIdx.y * 2 + 1) * C + c] = ss;
   }
 }
 
@@ -115,8 +115,8 @@ __global__ void bn_stats_nhwc_kernel(const T* __restrict__ x,
       s += tmp[j * CG + ci];
       ss += tmp[256 + j * CG + ci];
     }
-    atomicAdd(&sums[c], s);
-    atomicAdd(&sums[C + c], ss);
+    sums[((int64_t)blockIdx.y * 2 + 0) * C + c] = s;
+    sums[((int64_t)blockIdx.y * 2 + 1) * C + c] = ss;
   }
 }
 
@@ -170,8 +170,8 @@ __global__ void bn_stats_nhwc_vec_kernel(const T* __restrict__ x,
       }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sums[c0 + j], (double)s[j]);
-      atomicAdd(&sums[C + c0 + j], (double)ss[j]);
+      sums[((int64_t)blockIdx.y * 2 + 0) * C + c0 + j] = (double)s[j];
+      sums[((int64_t)blockIdx.y * 2 + 1) * C + c0 + j] = (double)ss[j];
     }
   }
 }
@@ -218,27 +218,33 @@ __global__ void bn_stats_nhwc_vec_f32_kernel(const float* __restrict__ x,
       }
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      atomicAdd(&sums[c0 + j], (double)s[j]);
-      atomicAdd(&sums[C + c0 + j], (double)ss[j]);
+      sums[((int64_t)blockIdx.y * 2 + 0) * C + c0 + j] = (double)s[j];
+      sums[((int64_t)blockIdx.y * 2 + 1) * C + c0 + j] = (double)ss[j];
     }
   }
 }
 
 // ---- finalize: mean/invstd + running-stat update ----------------------
-__global__ void bn_finalize_kernel(double* __restrict__ sums,
+// sums = per-(row-slice) partial slabs [S][2][C] (the reduce kernels
+// write plain stores — a C=64 layer at S~1500 slices would otherwise
+// serialize ~1500 f64 atomics per channel address)
+__global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
-                                   int64_t C, double count, float eps,
-                                   float momentum) {
+                                   int64_t C, int64_t S, double count,
+                                   float eps, float momentum) {
   const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const double m = sums[c] / count;
-  double var = sums[C + c] / count - m * m;
+  double s = 0.0, ss = 0.0;
+  for (int64_t b = 0; b < S; ++b) {
+    s += sums[(b * 2 + 0) * C + c];
+    ss += sums[(b * 2 + 1) * C + c];
+  }
+  const double m = s / count;
+  double var = ss / count - m * m;
   var = var < 0.0 ? 0.0 : var;
-  sums[c] = 0.0;      // leave the workspace zeroed for the next caller
-  sums[C + c] = 0.0;
   mean[c] = (float)m;
   invstd[c] = (float)rsqrt(var + (double)eps);
   if (running_mean != nullptr) {
@@ -555,8 +561,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
   __syncthreads();
   sdyx = block_reduce(sdyx, tmp, op, 0.0);
   if (threadIdx.x == 0) {
-    atomicAdd(&sums[c], sdy);
-    atomicAdd(&sums[C + c], sdyx);
+    sums[((int64_t)blockIdx.y * 2 + 0) * C + c] = sdy;
+    sums[((int64_t)blockIdx.y * 2 + 1) * C + c] = sdyx;
   }
 }
 
@@ -605,8 +611,8 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const T* __restrict__ dy,
       sdy += tmp[j * CG + ci];
       sdyx += tmp[256 + j * CG + ci];
     }
-    atomicAdd(&sums[c], sdy);
-    atomicAdd(&sums[C + c], sdyx);
+    sums[((int64_t)blockIdx.y * 2 + 0) * C + c] = sdy;
+    sums[((int64_t)blockIdx.y * 2 + 1) * C + c] = sdyx;
   }
 }
 
@@ -685,25 +691,27 @@ __global__ void bn_bwd_reduce_nhwc_vec_kernel(
       }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sums[c0 + j], (double)sdy[j]);
-      atomicAdd(&sums[C + c0 + j], (double)sdyx[j]);
+      sums[((int64_t)blockIdx.y * 2 + 0) * C + c0 + j] = (double)sdy[j];
+      sums[((int64_t)blockIdx.y * 2 + 1) * C + c0 + j] = (double)sdyx[j];
     }
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(double* __restrict__ sums,
+__global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ invstd,
                                        float* __restrict__ dgamma,
                                        float* __restrict__ dbeta,
                                        float* __restrict__ k,  // [3][C]
-                                       int64_t C, double count,
+                                       int64_t C, int64_t S, double count,
                                        int training) {
   const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const double sdy = sums[c], sdyx = sums[C + c];
-  sums[c] = 0.0;
-  sums[C + c] = 0.0;
+  double sdy = 0.0, sdyx = 0.0;
+  for (int64_t b = 0; b < S; ++b) {
+    sdy += sums[(b * 2 + 0) * C + c];
+    sdyx += sums[(b * 2 + 1) * C + c];
+  }
   dgamma[c] = (float)sdyx;
   dbeta[c] = (float)sdy;
   k[c] = gamma[c] * invstd[c];                       // k1
@@ -738,38 +746,55 @@ static inline int elementwise_grid(int64_t total, int block) {
   return (int)i64min(want > 0 ? want : 1, 256 * 8);
 }
 
+
+// Row-slice block count (gridDim.y) shared by the stats/bwd-reduce
+// kernels AND the workspace allocation in the bindings: the partial
+// slab is [S][2][C] doubles. Must mirror the launchers' dispatch.
+int64_t bn_reduce_gridS(int64_t N, int64_t C, int64_t HW, int nhwc,
+                        int elsize) {
+  if (nhwc) {
+    const int64_t rows = N * HW;
+    int64_t cblocks;
+    if (elsize == 2 && C % 8 == 0 && g_bn_variant == 2) {
+      const int64_t CG8 = i64min(C / 8, 64);
+      cblocks = (C / 8 + CG8 - 1) / CG8;
+    } else if (elsize == 4 && C % 4 == 0 && g_bn_variant == 2) {
+      const int64_t CG4 = i64min(C / 4, 64);
+      cblocks = (C / 4 + CG4 - 1) / CG4;
+    } else {
+      const int64_t CG = C >= 64 ? 64 : C;
+      cblocks = (C + CG - 1) / CG;
+    }
+    return i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
+  }
+  int64_t S = i64min((N * HW + 255) / 256, i64max(2048 / C, 1));
+  return i64max(S, 1);
+}
+
 template <typename T>
 void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
-                     int64_t HW, int nhwc, hipStream_t stream) {
+                     int64_t HW, int64_t S, int nhwc, hipStream_t stream) {
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
     if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
-      int64_t S = i64min(i64max(rows / 512, 1),
-                         i64max(2048 / cblocks, 1));
       hipLaunchKernelGGL((bn_stats_nhwc_vec_kernel<T>), dim3(cblocks, S),
                          dim3(block), 0, stream, x, sums, rows, C, CG8);
     } else if (sizeof(T) == 4 && C % 4 == 0 && g_bn_variant == 2) {
       const int CG4 = (int)i64min(C / 4, 64);
       const int64_t cblocks = (C / 4 + CG4 - 1) / CG4;
-      int64_t S = i64min(i64max(rows / 512, 1),
-                         i64max(2048 / cblocks, 1));
       hipLaunchKernelGGL(bn_stats_nhwc_vec_f32_kernel, dim3(cblocks, S),
                          dim3(block), 0, stream, (const float*)x, sums,
                          rows, C, CG4);
     } else {
       const int CG = C >= 64 ? 64 : (int)C;
       const int64_t cblocks = (C + CG - 1) / CG;
-      int64_t S = i64min(i64max(rows / 512, 1),
-                         i64max(2048 / cblocks, 1));
       hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
                          dim3(block), 0, stream, x, sums, rows, C, CG);
     }
   } else {
-    int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
-    S = i64max(S, 1);
     hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(C, S), dim3(block), 0,
                        stream, x, sums, N, C, HW);
   }
@@ -777,12 +802,13 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
 }
 
 void launch_bn_finalize(double* sums, float* mean, float* invstd,
-                        float* rm, float* rv, int64_t C, double count,
+                        float* rm, float* rv, int64_t C, int64_t S,
+                        double count,
                         float eps, float momentum, hipStream_t stream) {
   const int block = 256;
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + block - 1) / block),
                      dim3(block), 0, stream, sums, mean, invstd, rm, rv, C,
-                     count, eps, momentum);
+                     S, count, eps, momentum);
   HIP_CHECK_LAST();
 }
 
@@ -855,7 +881,7 @@ template <typename T>
 void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
                           const float* mean, const float* invstd,
                           double* sums, const unsigned char* msk,
-                          int64_t N, int64_t C, int64_t HW,
+                          int64_t N, int64_t C, int64_t HW, int64_t S,
                           int act, int nhwc, hipStream_t stream) {
   const int block = 256;
   if (nhwc) {
@@ -864,8 +890,6 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
     if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
       const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
-      int64_t S = i64min(i64max(rows / 512, 1),
-                         i64max(2048 / cblocks, 1));
 #define CASE(ACT, MSK)                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_vec_kernel<T, ACT, MSK>),      \
                          dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
@@ -877,8 +901,6 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
     } else {
       const int CG = C >= 64 ? 64 : (int)C;
       const int64_t cblocks = (C + CG - 1) / CG;
-      int64_t S = i64min(i64max(rows / 512, 1),
-                         i64max(2048 / cblocks, 1));
 #define CASE(ACT, MSK)                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT, MSK>),          \
                          dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
@@ -889,8 +911,6 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
 #undef CASE
     }
   } else {
-    int64_t S = i64min((N * HW + block - 1) / block, i64max(2048 / C, 1));
-    S = i64max(S, 1);
 #define CASE(ACT)                                                          \
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, ACT>), dim3(C, S),         \
                        dim3(block), 0, stream, dy, y, x, mean, invstd,     \
@@ -905,12 +925,12 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
 
 void launch_bn_bwd_finalize(double* sums, const float* gamma,
                             const float* invstd, float* dgamma, float* dbeta,
-                            float* k, int64_t C, double count, int training,
-                            hipStream_t stream) {
+                            float* k, int64_t C, int64_t S, double count,
+                            int training, hipStream_t stream) {
   const int block = 256;
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + block - 1) / block),
                      dim3(block), 0, stream, sums, gamma, invstd, dgamma,
-                     dbeta, k, C, count, training);
+                     dbeta, k, C, S, count, training);
   HIP_CHECK_LAST();
 }
 
@@ -987,7 +1007,7 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
 
 #define INSTANTIATE(T)                                                        \
   template void launch_bn_stats<T>(const T*, double*, int64_t, int64_t,       \
-                                   int64_t, int, hipStream_t);                \
+                                   int64_t, int64_t, int, hipStream_t);       \
   template bool launch_bn_apply<T>(const T*, const T*, T*, const float*,      \
                                    const float*, const float*, const float*,  \
                                    unsigned char*,                            \
@@ -996,8 +1016,8 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
   template void launch_bn_bwd_reduce<T>(const T*, const T*, const T*,         \
                                         const float*, const float*, double*,  \
                                         const unsigned char*,                 \
-                                        int64_t, int64_t, int64_t, int, int,  \
-                                        hipStream_t);                         \
+                                        int64_t, int64_t, int64_t, int64_t,   \
+                                        int, int, hipStream_t);               \
   template void launch_bn_bwd_dx<T>(const T*, const T*, const T*,             \
                                     const float*, const float*, const float*, \
                                     T*, T*, const unsigned char*,             \
