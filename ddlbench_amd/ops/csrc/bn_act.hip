@@ -228,6 +228,7 @@ __global__ void bn_stats_nhwc_vec_f32_kernel(const float* __restrict__ x,
 // sums = per-(row-slice) partial slabs [S][2][C] (the reduce kernels
 // write plain stores — a C=64 layer at S~1500 slices would otherwise
 // serialize ~1500 f64 atomics per channel address)
+// one block per channel; 256 threads block-reduce the S slabs
 __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
@@ -235,13 +236,18 @@ __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ running_var,
                                    int64_t C, int64_t S, double count,
                                    float eps, float momentum) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ double tmp[8];
+  const int64_t c = blockIdx.x;
   double s = 0.0, ss = 0.0;
-  for (int64_t b = 0; b < S; ++b) {
+  for (int64_t b = threadIdx.x; b < S; b += blockDim.x) {
     s += sums[(b * 2 + 0) * C + c];
     ss += sums[(b * 2 + 1) * C + c];
   }
+  auto op = [](double v) { return wave_reduce_sum(v); };
+  s = block_reduce(s, tmp, op, 0.0);
+  __syncthreads();
+  ss = block_reduce(ss, tmp, op, 0.0);
+  if (threadIdx.x != 0) return;
   const double m = s / count;
   double var = ss / count - m * m;
   var = var < 0.0 ? 0.0 : var;
@@ -705,13 +711,18 @@ __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
                                        float* __restrict__ k,  // [3][C]
                                        int64_t C, int64_t S, double count,
                                        int training) {
-  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ double tmp[8];
+  const int64_t c = blockIdx.x;
   double sdy = 0.0, sdyx = 0.0;
-  for (int64_t b = 0; b < S; ++b) {
+  for (int64_t b = threadIdx.x; b < S; b += blockDim.x) {
     sdy += sums[(b * 2 + 0) * C + c];
     sdyx += sums[(b * 2 + 1) * C + c];
   }
+  auto op = [](double v) { return wave_reduce_sum(v); };
+  sdy = block_reduce(sdy, tmp, op, 0.0);
+  __syncthreads();
+  sdyx = block_reduce(sdyx, tmp, op, 0.0);
+  if (threadIdx.x != 0) return;
   dgamma[c] = (float)sdyx;
   dbeta[c] = (float)sdy;
   k[c] = gamma[c] * invstd[c];                       // k1
@@ -806,7 +817,7 @@ void launch_bn_finalize(double* sums, float* mean, float* invstd,
                         double count,
                         float eps, float momentum, hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + block - 1) / block),
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((unsigned)C),
                      dim3(block), 0, stream, sums, mean, invstd, rm, rv, C,
                      S, count, eps, momentum);
   HIP_CHECK_LAST();
@@ -928,7 +939,7 @@ void launch_bn_bwd_finalize(double* sums, const float* gamma,
                             float* k, int64_t C, int64_t S, double count,
                             int training, hipStream_t stream) {
   const int block = 256;
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + block - 1) / block),
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((unsigned)C),
                      dim3(block), 0, stream, sums, gamma, invstd, dgamma,
                      dbeta, k, C, S, count, training);
   HIP_CHECK_LAST();
