@@ -58,7 +58,8 @@ void launch_ce_bwd(const T*, const int64_t*, const float*, const float*, T*,
 void launch_conv_igemm(const void*, const void*, void*, const void*, int,
                        int, int, int, int, int, int, int, int, int, int,
                        int, hipStream_t);
-void launch_conv_wgrad(const void*, const void*, float*, const void*, int,
+void launch_conv_wgrad(const void*, const void*, float*, float*, long,
+                       const void*, int,
                        int, int, int, int, int, int, int, int, int, int,
                        hipStream_t);
 
@@ -406,8 +407,19 @@ torch::Tensor conv_igemm_wgrad(torch::Tensor x, torch::Tensor dy,
               "conv_wgrad needs C %% 8 == 0 and K %% 8 == 0");
   auto dw = torch::zeros({K, R * S * C}, x.options().dtype(at::kFloat));
   auto zero = torch::zeros({16}, x.options());
+  // split-P partial-slab workspace (cached per device, 32 MB fp32)
+  static std::unordered_map<int, torch::Tensor> wgws;
+  const int devi = (int)x.get_device();
+  auto wit = wgws.find(devi);
+  if (wit == wgws.end()) {
+    wgws[devi] = torch::empty({8 << 20},
+                              x.options().dtype(at::kFloat));
+    wit = wgws.find(devi);
+  }
   launch_conv_wgrad(x.data_ptr(), dy.data_ptr(),
-                    dw.data_ptr<float>(), zero.data_ptr(), N, H, W, C, K,
+                    dw.data_ptr<float>(), wit->second.data_ptr<float>(),
+                    (long)wit->second.numel(), zero.data_ptr(),
+                    N, H, W, C, K,
                     OH, OW, (int)R, (int)S, (int)stride, (int)pad,
                     cur_stream());
   return dw;

@@ -1,28 +1,44 @@
 // NHWC bf16 convolution weight-grad on MFMA (gfx950) — completes the
-// native conv triple (fprop/dgrad in conv_mfma.hip).
+// native conv triple (fprop/dgrad in conv_mfma*.hip).
 //
 //   dw[k][r,s,c] = sum_{n,oh,ow} dy[n,oh,ow,k] * x[n, oh*st+r-p, ow*st+s-p, c]
 //
-// GEMM view: C[k][rsc] = A[k][p] * B[p][rsc] with the reduction over
-// output pixels p. Both operands are p-major in memory (dy rows are
-// [p][K], im2col rows are [p][rsc]), i.e. TRANSPOSED from the fragment
-// layout, so both tiles stage into LDS row-major-in-p (16-byte
-// global_load_lds chunks, the same XOR chunk swizzle as conv_mfma) and
-// fragments gather with per-element transposed LDS reads. Split-P grid
-// accumulates fp32 partials with atomics.
+// GEMM view: C[k][rsc] = A[k][p] * B[p][rsc], reduction over output
+// pixels p. Both operands are p-major in memory (transposed from the
+// MFMA fragment layout), so tiles stage into a ds_read_b64_tr_b16 image
+// and fragments come out of the HARDWARE TRANSPOSE READ — one b64-class
+// LDS instruction per 4 fragment elements instead of the per-element
+// scalar gathers of the round-1 kernel (its 2.3x-vs-MIOpen gap).
 //
-// Correctness-first structure (one LDS buffer pair, scalar transposed
-// reads); the ds_read_b64_tr_b16 image is the planned next step
-// (docs/ROADMAP.md). Constraints: bf16, K % 8 == 0, C % 8 == 0.
+// Structure (v2, default):
+//   * 64k x 64rsc tile, BP=64 pixels per step, 256 threads (4 waves,
+//     one 16-rsc column group each).
+//   * tr image per operand: 4 column groups x [two p-parity
+//     sub-images] x [p/8][4][16] halfword blocks; glds stages it with
+//     lane-linear 16-B chunks whose SOURCE index is decoded per slot
+//     (cdna_hip_programming.md T10: lane l of a 16-lane group reads
+//     column l&15 of a [4][16] block at lane-linear 8-B addresses).
+//   * 3-buffer LDS ring (48 KiB), staging 2 steps ahead, counted
+//     s_waitcnt vmcnt(4) per step, raw s_barrier (never __syncthreads
+//     with glds in flight).
+//   * split-P partial slabs [split][K][RSC] fp32 + a combine kernel —
+//     no global atomics (a C64K64R1 layer would serialize ~2048 f64
+//     atomic adds per dw element).
+// DDLB_WGRAD_V2=0 falls back to the round-1 scalar-gather kernel.
+//
+// Constraints: bf16, K % 8 == 0, C % 8 == 0.
 
 #include "common.h"
 #include <stdint.h>
 #include <stdexcept>
+#include <stdlib.h>
 #include <string>
 
 using bf16 = __hip_bfloat16;
 typedef __attribute__((ext_vector_type(8))) short bf16x8w;
+typedef __attribute__((ext_vector_type(4))) short bf16x4w;
 typedef __attribute__((ext_vector_type(4))) float f32x4w;
+typedef __attribute__((address_space(3))) const short* lds_cptr;
 
 #define WG_BKP 64     // p rows per step
 #define WG_TK 64      // k tile
@@ -32,7 +48,8 @@ typedef __attribute__((ext_vector_type(4))) float f32x4w;
 struct WgradParams {
   const bf16* x;      // (N,H,W,C) memory
   const bf16* dy;     // (N,OH,OW,K) memory
-  float* dw;          // (K, R*S*C) fp32, zero-initialized
+  float* dw;          // (K, R*S*C) fp32 output
+  float* part;        // [split][K][RSC] fp32 partial slabs (split>1)
   const bf16* zero;
   int N, H, W, C, K, OH, OW, R, S, stride, pad;
   long P;             // N*OH*OW
@@ -40,6 +57,9 @@ struct WgradParams {
   int split_p;        // p-chunks per (k,rsc) tile
 };
 
+// ---------------------------------------------------------------------
+// v1 kernel (round 1): scalar transposed gathers. Kept for A/B and as
+// the DDLB_WGRAD_V2=0 fallback.
 __global__ __launch_bounds__(WG_THREADS, 2)
 void conv_wgrad_kernel(WgradParams q) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -60,12 +80,8 @@ void conv_wgrad_kernel(WgradParams q) {
   const long p_begin = ps_idx * p_per;
   const long p_end = i64min(p_begin + p_per, q.P);
 
-  // staging slots: 512 chunks per tile, 2 per thread per tile;
-  // slot (prow = ca>>3, cg_store = ca&7) holds source chunk
-  // cg = cg_store ^ (prow & 7)  (16B chunks stay contiguous)
   const int nsteps = (int)((p_end - p_begin + WG_BKP - 1) / WG_BKP);
 
-  // wave tile: 64k x 16rsc (4 waves side by side in rsc)
   const int wn = wid * 16;
   f32x4w acc[4];
 #pragma unroll
@@ -73,7 +89,6 @@ void conv_wgrad_kernel(WgradParams q) {
 
   for (int t = 0; t < nsteps; ++t) {
     const long p0 = p_begin + (long)t * WG_BKP;
-    // ---- stage A (dy) and B (im2col x) --------------------------------
 #pragma unroll
     for (int l = 0; l < 2; ++l) {
       const int ca = l * WG_THREADS + tid;
@@ -117,7 +132,6 @@ void conv_wgrad_kernel(WgradParams q) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
-    // ---- transposed fragment reads + MFMA -----------------------------
     // element (p, col) lives at p*64 + ((col>>3)^(p&7))*8 + (col&7)
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -148,8 +162,7 @@ void conv_wgrad_kernel(WgradParams q) {
     __syncthreads();
   }
 
-  // ---- fp32 atomic accumulation into dw -------------------------------
-  // D[row = (lane>>4)*4 + reg][col = lane&15]; row is the k index
+  // fp32 atomic accumulation into dw (v1 path keeps atomics)
 #pragma unroll
   for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
@@ -162,14 +175,230 @@ void conv_wgrad_kernel(WgradParams q) {
   }
 }
 
+// ---------------------------------------------------------------------
+// v2: hardware-transpose-read pipeline.
+namespace {
+
+// tr image geometry (per operand, per buffer): 4 column groups g of 16,
+// each a pair of p-parity sub-images of [BP/8][4][16] halfword blocks:
+//   half(p, ch) = g*BP*16 + ((p>>2)&1)*BP*8 + (p>>3)*64 + (p&3)*16
+//                 + (ch&15)
+// A 16-B staging chunk d in [0, 512) covers (p, ch0=8*h8):
+DEV void wg_decode_chunk(int d, int& p, int& ch) {
+  const int g = d >> 7;
+  const int r = d & 127;
+  const int phalf = r >> 6;
+  const int r2 = r & 63;
+  p = (r2 >> 3) * 8 + phalf * 4 + ((r2 >> 1) & 3);
+  ch = g * 16 + (r2 & 1) * 8;
+}
+
+DEV bf16x8w wg_tr_frag(const short* img, int group, int ks, int lane) {
+  // fragment (m/n-col = lane&15, p = ks*32 + (lane>>4)*8 + e):
+  // two tr reads (p parity halves) at lane-linear 8-B addresses
+  bf16x4w lo, hi;
+  lds_cptr base = (lds_cptr)(img + group * (WG_BKP * 16) + ks * 4 * 64 +
+                             lane * 4);
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+               "ds_read_b64_tr_b16 %1, %2 offset:%3"
+               : "=&v"(lo), "=&v"(hi)
+               : "v"(base), "i"(WG_BKP * 8 * 2));
+  bf16x8w f;
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    f[e] = lo[e];
+    f[4 + e] = hi[e];
+  }
+  return f;
+}
+
+__global__ __launch_bounds__(WG_THREADS, 2)
+void conv_wgrad2_kernel(WgradParams q) {
+  // 3-buffer ring: [buf][A 64x64 | B 64x64] bf16
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* lds = reinterpret_cast<short*>(smem);
+  auto aimg = [&](int buf) { return lds + buf * 2 * WG_BKP * 64; };
+  auto bimg = [&](int buf) { return aimg(buf) + WG_BKP * 64; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  const int nk = (q.K + WG_TK - 1) / WG_TK;
+  const int nr = (int)((q.RSC + WG_TR - 1) / WG_TR);
+  const long block = blockIdx.x;
+  const long kb = (block % nk) * WG_TK;
+  const long rb = ((block / nk) % nr) * WG_TR;
+  const int ps_idx = (int)(block / ((long)nk * nr));
+  const long p_per = (q.P + q.split_p - 1) / q.split_p;
+  const long p_begin = ps_idx * p_per;
+  const long p_end = i64min(p_begin + p_per, q.P);
+  const int nsteps = (int)((p_end - p_begin + WG_BKP - 1) / WG_BKP);
+
+  // ---- staging slots: 2 chunks per thread per operand ------------------
+  // A (dy): source advances linearly by BP*K per step.
+  // B (x im2col): fixed (r,s,c0) per slot; (n,oh,ow) advances by BP
+  // rows per step with carries.
+  int a_p[2], a_ch[2];
+  const bf16* a_src[2];
+  bool a_chok[2];
+  int b_r[2], b_s[2], b_c0[2];
+  int b_n[2], b_oh[2], b_ow[2];
+  bool b_rscok[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int d = j * WG_THREADS + tid;
+    wg_decode_chunk(d, a_p[j], a_ch[j]);
+    const long rsc = rb + a_ch[j];
+    b_rscok[j] = rsc < q.RSC;
+    const long rr = b_rscok[j] ? rsc : 0;
+    b_c0[j] = (int)(rr % q.C);
+    const int rs = (int)(rr / q.C);
+    b_r[j] = rs / q.S;
+    b_s[j] = rs - b_r[j] * q.S;
+    a_chok[j] = kb + a_ch[j] < q.K;
+    a_src[j] = q.dy + (p_begin + a_p[j]) * q.K + kb + a_ch[j];
+    // decompose pixel p_begin + p once
+    const long pp = p_begin + a_p[j];
+    const int ohw = q.OH * q.OW;
+    const long ppc = pp < q.P ? pp : 0;
+    b_n[j] = (int)(ppc / ohw);
+    const int rem = (int)(ppc - (long)b_n[j] * ohw);
+    b_oh[j] = rem / q.OW;
+    b_ow[j] = rem - b_oh[j] * q.OW;
+  }
+  long a_pabs[2] = {p_begin + a_p[0], p_begin + a_p[1]};
+
+  auto stage = [&](int buf, int t) {
+    short* la = aimg(buf);
+    short* lb = bimg(buf);
+    const long plim = p_end;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int d = j * WG_THREADS + tid;
+      const bf16* src =
+          (a_chok[j] && a_pabs[j] + (long)t * WG_BKP < plim)
+              ? a_src[j] + (long)t * WG_BKP * q.K
+              : q.zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(la + d * 8), 16, 0, 0);
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int d = j * WG_THREADS + tid;
+      // advance (n,oh,ow) to step t lazily: ow' = ow + t*BP with carry
+      // would need history; instead recompute from the running state —
+      // the stage calls are strictly t = 0,1,2,... so the running state
+      // IS step t's state; advance after use.
+      const bf16* src = q.zero;
+      if (b_rscok[j] && a_pabs[j] + (long)t * WG_BKP < plim) {
+        const int ih = b_oh[j] * q.stride + b_r[j] - q.pad;
+        const int iw = b_ow[j] * q.stride + b_s[j] - q.pad;
+        if (ih >= 0 && ih < q.H && iw >= 0 && iw < q.W)
+          src = q.x + (((long)b_n[j] * q.H + ih) * q.W + iw) * q.C +
+                b_c0[j];
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(lb + d * 8), 16, 0, 0);
+      // advance the pixel coords by BP rows (carry chain)
+      b_ow[j] += WG_BKP;
+      while (b_ow[j] >= q.OW) {
+        b_ow[j] -= q.OW;
+        if (++b_oh[j] == q.OH) {
+          b_oh[j] = 0;
+          ++b_n[j];
+        }
+      }
+    }
+  };
+
+  f32x4w acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  // ---- prologue: stage steps 0 and 1 -----------------------------------
+  stage(0, 0);
+  if (nsteps > 1) {
+    stage(1, 1);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // step 0 landed
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < nsteps; ++t) {
+    const short* la = aimg(t % 3);
+    const short* lb = bimg(t % 3);
+    // fragments via hardware transpose read
+    bf16x8w af[2][4], bf_[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf_[ks] = wg_tr_frag(lb, wid, ks, lane);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        af[ks][mf] = wg_tr_frag(la, mf, ks, lane);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);  // keep MFMAs behind the wait
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[ks][mf], bf_[ks], acc[mf], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    if (t + 2 < nsteps) {
+      stage((t + 2) % 3, t + 2);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // ---- write the tile: partial slab (split>1) or dw directly ----------
+  float* out = (q.split_p > 1)
+                   ? q.part + (long)ps_idx * q.K * q.RSC
+                   : q.dw;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const long k = kb + mf * 16 + (lane >> 4) * 4 + reg;
+      const long rsc = rb + wid * 16 + (lane & 15);
+      if (k < q.K && rsc < q.RSC)
+        out[k * q.RSC + rsc] = acc[mf][reg];
+    }
+  }
+}
+
+__global__ void wgrad_combine_kernel(const float* __restrict__ part,
+                                     float* __restrict__ dw, long total,
+                                     long stride_elems, int split) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long gstride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < total; i += gstride) {
+    float s = 0.f;
+    for (int ps = 0; ps < split; ++ps) s += part[ps * stride_elems + i];
+    dw[i] = s;
+  }
+}
+
+}  // namespace
+
 void launch_conv_wgrad(const void* x, const void* dy, float* dw,
-                       const void* zero, int N, int H, int W, int C, int K,
+                       float* part_ws, long part_cap, const void* zero,
+                       int N, int H, int W, int C, int K,
                        int OH, int OW, int R, int S, int stride, int pad,
                        hipStream_t stream) {
   WgradParams q;
   q.x = (const bf16*)x;
   q.dy = (const bf16*)dy;
   q.dw = dw;
+  q.part = part_ws;
   q.zero = (const bf16*)zero;
   q.N = N; q.H = H; q.W = W; q.C = C; q.K = K; q.OH = OH; q.OW = OW;
   q.R = R; q.S = S; q.stride = stride; q.pad = pad;
@@ -177,6 +406,31 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
   q.RSC = (long)R * S * C;
   const long nk = (K + WG_TK - 1) / WG_TK;
   const long nr = (q.RSC + WG_TR - 1) / WG_TR;
+
+  const char* v2e = getenv("DDLB_WGRAD_V2");
+  const bool use_v2 = !(v2e && v2e[0] == '0') && part_ws != nullptr;
+
+  if (use_v2) {
+    long split = 2048 / i64max(nk * nr, 1);
+    split = i64max(i64min(split, (q.P + WG_BKP - 1) / WG_BKP), 1);
+    if (split > 1)
+      split = i64min(split, part_cap / i64max(q.K * q.RSC, 1));
+    split = i64max(split, 1);
+    q.split_p = (int)split;
+    const size_t lds2 = 3 * 2 * WG_BKP * 64 * sizeof(bf16);
+    hipLaunchKernelGGL(conv_wgrad2_kernel,
+                       dim3((unsigned)(nk * nr * q.split_p)),
+                       dim3(WG_THREADS), lds2, stream, q);
+    if (q.split_p > 1) {
+      const long total = q.K * q.RSC;
+      const int blocks = (int)i64min((total + 255) / 256, 2048);
+      hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
+                         stream, part_ws, dw, total, total, q.split_p);
+    }
+    HIP_CHECK_LAST();
+    return;
+  }
+
   long split = 2048 / i64max(nk * nr, 1);
   split = i64max(i64min(split, (q.P + WG_BKP - 1) / WG_BKP), 1);
   q.split_p = (int)split;
