@@ -102,12 +102,20 @@ def main():
             [0, 0], 1, [True, False, False]), args.iters, args.warmup)
         res["miopen_dgrad_ms"] = round(t * 1e3, 3)
         res["miopen_dgrad_tf"] = round(flops / t / 1e12, 1)
-        # wgrad
+        # wgrad: v2 (tr_b16 pipeline) then v1 (scalar gathers)
+        os.environ["DDLB_WGRAD_V2"] = "1"
+        t = bench_op(lambda: ext.conv_igemm_wgrad(x, dy, R, R, stride,
+                                                  pad),
+                     args.iters, args.warmup)
+        res["mfma2_wgrad_ms"] = round(t * 1e3, 3)
+        res["mfma2_wgrad_tf"] = round(flops / t / 1e12, 1)
+        os.environ["DDLB_WGRAD_V2"] = "0"
         t = bench_op(lambda: ext.conv_igemm_wgrad(x, dy, R, R, stride,
                                                   pad),
                      args.iters, args.warmup)
         res["mfma_wgrad_ms"] = round(t * 1e3, 3)
         res["mfma_wgrad_tf"] = round(flops / t / 1e12, 1)
+        os.environ["DDLB_WGRAD_V2"] = "1"
         t = bench_op(lambda: torch.ops.aten.convolution_backward(
             dy, x, w, None, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1, [False, True, False]), args.iters, args.warmup)
