@@ -179,6 +179,16 @@ void conv_wgrad_kernel(WgradParams q) {
 // v2: hardware-transpose-read pipeline.
 namespace {
 
+template <int N> DEV void wait_step_vmcnt() {
+  static_assert(N >= 0 && N <= 8, "vmcnt");
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+}
+
 // tr image geometry (per operand, per buffer): 4 column groups g of 16,
 // each a pair of p-parity sub-images of [BP/8][4][16] halfword blocks:
 //   half(p, ch) = g*BP*16 + ((p>>2)&1)*BP*8 + (p>>3)*64 + (p&3)*16
@@ -212,69 +222,87 @@ DEV bf16x8w wg_tr_frag(const short* img, int group, int ks, int lane) {
   return f;
 }
 
+// TKt x TRt dw tile, WR x WC waves. The 128x128 tile halves both
+// operands' cross-tile re-staging (the 64x64 tile is staging-BW-bound:
+// A re-reads x nr, B x nk).
+template <int TKt, int TRt, int WR, int WC>
 __global__ __launch_bounds__(WG_THREADS, 2)
 void conv_wgrad2_kernel(WgradParams q) {
-  // 3-buffer ring: [buf][A 64x64 | B 64x64] bf16
+  constexpr int GA = TKt / 32;       // A glds chunks per thread per step
+  constexpr int GB = TRt / 32;       // B
+  constexpr int MFS = TKt / WR / 16; // A fragments per wave
+  constexpr int NFS = TRt / WC / 16; // B fragments per wave
+  static_assert(WR * WC == 4, "4 waves");
+  // 3-buffer ring: [buf][A TKtx64 | B TRtx64] bf16
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* lds = reinterpret_cast<short*>(smem);
-  auto aimg = [&](int buf) { return lds + buf * 2 * WG_BKP * 64; };
-  auto bimg = [&](int buf) { return aimg(buf) + WG_BKP * 64; };
+  auto aimg = [&](int buf) { return lds + buf * WG_BKP * (TKt + TRt); };
+  auto bimg = [&](int buf) { return aimg(buf) + WG_BKP * TKt; };
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
-  const int nk = (q.K + WG_TK - 1) / WG_TK;
-  const int nr = (int)((q.RSC + WG_TR - 1) / WG_TR);
+  const int nk = (q.K + TKt - 1) / TKt;
+  const int nr = (int)((q.RSC + TRt - 1) / TRt);
   const long block = blockIdx.x;
-  const long kb = (block % nk) * WG_TK;
-  const long rb = ((block / nk) % nr) * WG_TR;
+  const long kb = (block % nk) * TKt;
+  const long rb = ((block / nk) % nr) * TRt;
   const int ps_idx = (int)(block / ((long)nk * nr));
   const long p_per = (q.P + q.split_p - 1) / q.split_p;
   const long p_begin = ps_idx * p_per;
   const long p_end = i64min(p_begin + p_per, q.P);
   const int nsteps = (int)((p_end - p_begin + WG_BKP - 1) / WG_BKP);
 
-  // ---- staging slots: 2 chunks per thread per operand ------------------
+  // ---- staging slots: GA/GB chunks per thread per operand --------------
   // A (dy): source advances linearly by BP*K per step.
   // B (x im2col): fixed (r,s,c0) per slot; (n,oh,ow) advances by BP
   // rows per step with carries.
-  int a_p[2], a_ch[2];
-  const bf16* a_src[2];
-  bool a_chok[2];
-  int b_r[2], b_s[2], b_c0[2];
-  int b_n[2], b_oh[2], b_ow[2];
-  bool b_rscok[2];
+  int a_p[GA], a_ch[GA];
+  const bf16* a_src[GA];
+  bool a_chok[GA];
+  long a_pabs[GA];
+  int bp_[GB], b_r[GB], b_s[GB], b_c0[GB];
+  int b_n[GB], b_oh[GB], b_ow[GB];
+  bool b_rscok[GB];
+  long b_pabs[GB];
 #pragma unroll
-  for (int j = 0; j < 2; ++j) {
+  for (int j = 0; j < GA; ++j) {
     const int d = j * WG_THREADS + tid;
     wg_decode_chunk(d, a_p[j], a_ch[j]);
-    const long rsc = rb + a_ch[j];
+    a_chok[j] = kb + a_ch[j] < q.K;
+    a_src[j] = q.dy + (p_begin + a_p[j]) * q.K + kb + a_ch[j];
+    a_pabs[j] = p_begin + a_p[j];
+  }
+#pragma unroll
+  for (int j = 0; j < GB; ++j) {
+    const int d = j * WG_THREADS + tid;
+    int bch;
+    wg_decode_chunk(d, bp_[j], bch);
+    const long rsc = rb + bch;
     b_rscok[j] = rsc < q.RSC;
     const long rr = b_rscok[j] ? rsc : 0;
     b_c0[j] = (int)(rr % q.C);
     const int rs = (int)(rr / q.C);
     b_r[j] = rs / q.S;
     b_s[j] = rs - b_r[j] * q.S;
-    a_chok[j] = kb + a_ch[j] < q.K;
-    a_src[j] = q.dy + (p_begin + a_p[j]) * q.K + kb + a_ch[j];
     // decompose pixel p_begin + p once
-    const long pp = p_begin + a_p[j];
+    const long pp = p_begin + bp_[j];
     const int ohw = q.OH * q.OW;
     const long ppc = pp < q.P ? pp : 0;
     b_n[j] = (int)(ppc / ohw);
     const int rem = (int)(ppc - (long)b_n[j] * ohw);
     b_oh[j] = rem / q.OW;
     b_ow[j] = rem - b_oh[j] * q.OW;
+    b_pabs[j] = pp;
   }
-  long a_pabs[2] = {p_begin + a_p[0], p_begin + a_p[1]};
 
   auto stage = [&](int buf, int t) {
     short* la = aimg(buf);
     short* lb = bimg(buf);
     const long plim = p_end;
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < GA; ++j) {
       const int d = j * WG_THREADS + tid;
       const bf16* src =
           (a_chok[j] && a_pabs[j] + (long)t * WG_BKP < plim)
@@ -285,14 +313,14 @@ void conv_wgrad2_kernel(WgradParams q) {
           (__attribute__((address_space(3))) void*)(la + d * 8), 16, 0, 0);
     }
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < GB; ++j) {
       const int d = j * WG_THREADS + tid;
       // advance (n,oh,ow) to step t lazily: ow' = ow + t*BP with carry
       // would need history; instead recompute from the running state —
       // the stage calls are strictly t = 0,1,2,... so the running state
       // IS step t's state; advance after use.
       const bf16* src = q.zero;
-      if (b_rscok[j] && a_pabs[j] + (long)t * WG_BKP < plim) {
+      if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim) {
         const int ih = b_oh[j] * q.stride + b_r[j] - q.pad;
         const int iw = b_ow[j] * q.stride + b_s[j] - q.pad;
         if (ih >= 0 && ih < q.H && iw >= 0 && iw < q.W)
@@ -314,47 +342,54 @@ void conv_wgrad2_kernel(WgradParams q) {
     }
   };
 
-  f32x4w acc[4];
+  const int wk = wid / WC;
+  const int wc = wid % WC;
+  f32x4w acc[MFS][NFS];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < MFS; ++i)
+#pragma unroll
+    for (int j = 0; j < NFS; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // ---- prologue: stage steps 0 and 1 -----------------------------------
   stage(0, 0);
   if (nsteps > 1) {
     stage(1, 1);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // step 0 landed
+    wait_step_vmcnt<GA + GB>();  // step 0 landed, step 1 in flight
   } else {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    wait_step_vmcnt<0>();
   }
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < nsteps; ++t) {
     const short* la = aimg(t % 3);
     const short* lb = bimg(t % 3);
-    // fragments via hardware transpose read
-    bf16x8w af[2][4], bf_[2];
+    // fragments via hardware transpose read; process the two 32-p
+    // halves (ks) back to back
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      bf_[ks] = wg_tr_frag(lb, wid, ks, lane);
+      bf16x8w af[MFS], bf_[NFS];
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
-        af[ks][mf] = wg_tr_frag(la, mf, ks, lane);
+      for (int nf = 0; nf < NFS; ++nf)
+        bf_[nf] = wg_tr_frag(lb, wc * NFS + nf, ks, lane);
+#pragma unroll
+      for (int mf = 0; mf < MFS; ++mf)
+        af[mf] = wg_tr_frag(la, wk * MFS + mf, ks, lane);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);  // keep MFMAs behind the wait
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < MFS; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NFS; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mf], bf_[nf], acc[mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);  // keep MFMAs behind the wait
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
-        acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[ks][mf], bf_[ks], acc[mf], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
     if (t + 2 < nsteps) {
       stage((t + 2) % 3, t + 2);
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      wait_step_vmcnt<GA + GB>();
     } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      wait_step_vmcnt<0>();
     }
     __builtin_amdgcn_s_barrier();
   }
@@ -364,26 +399,36 @@ void conv_wgrad2_kernel(WgradParams q) {
                    ? q.part + (long)ps_idx * q.K * q.RSC
                    : q.dw;
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < MFS; ++mf) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      const long k = kb + mf * 16 + (lane >> 4) * 4 + reg;
-      const long rsc = rb + wid * 16 + (lane & 15);
-      if (k < q.K && rsc < q.RSC)
-        out[k * q.RSC + rsc] = acc[mf][reg];
+      const long k = kb + (wk * MFS + mf) * 16 + (lane >> 4) * 4 + reg;
+      if (k >= q.K) continue;
+#pragma unroll
+      for (int nf = 0; nf < NFS; ++nf) {
+        const long rsc = rb + (wc * NFS + nf) * 16 + (lane & 15);
+        if (rsc < q.RSC) out[k * q.RSC + rsc] = acc[mf][nf][reg];
+      }
     }
   }
 }
 
+// 2D combine: blockIdx.y covers 64-slab chunks (atomics per element are
+// <= ceil(split/64)-way) so small dw tensors still parallelize
 __global__ void wgrad_combine_kernel(const float* __restrict__ part,
                                      float* __restrict__ dw, long total,
                                      long stride_elems, int split) {
+  const int ps0 = blockIdx.y * 64;
+  const int ps1 = min(ps0 + 64, split);
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long gstride = (long)gridDim.x * blockDim.x;
   for (long i = i0; i < total; i += gstride) {
     float s = 0.f;
-    for (int ps = 0; ps < split; ++ps) s += part[ps * stride_elems + i];
-    dw[i] = s;
+    for (int ps = ps0; ps < ps1; ++ps) s += part[ps * stride_elems + i];
+    if (gridDim.y > 1)
+      atomicAdd(&dw[i], s);
+    else
+      dw[i] = s;
   }
 }
 
@@ -411,20 +456,32 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
   const bool use_v2 = !(v2e && v2e[0] == '0') && part_ws != nullptr;
 
   if (use_v2) {
-    long split = 2048 / i64max(nk * nr, 1);
+    const bool big = (K >= 128 && q.RSC >= 128);
+    const long TKt = big ? 128 : 64, TRt = big ? 128 : 64;
+    const long nk2 = (K + TKt - 1) / TKt;
+    const long nr2 = (q.RSC + TRt - 1) / TRt;
+    long split = 2048 / i64max(nk2 * nr2, 1);
     split = i64max(i64min(split, (q.P + WG_BKP - 1) / WG_BKP), 1);
+    split = i64min(split, 256);  // combine traffic cap
     if (split > 1)
       split = i64min(split, part_cap / i64max(q.K * q.RSC, 1));
     split = i64max(split, 1);
     q.split_p = (int)split;
-    const size_t lds2 = 3 * 2 * WG_BKP * 64 * sizeof(bf16);
-    hipLaunchKernelGGL(conv_wgrad2_kernel,
-                       dim3((unsigned)(nk * nr * q.split_p)),
-                       dim3(WG_THREADS), lds2, stream, q);
+    const size_t lds2 = 3 * WG_BKP * (TKt + TRt) * sizeof(bf16);
+    if (big)
+      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 2>),
+                         dim3((unsigned)(nk2 * nr2 * q.split_p)),
+                         dim3(WG_THREADS), lds2, stream, q);
+    else
+      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4>),
+                         dim3((unsigned)(nk2 * nr2 * q.split_p)),
+                         dim3(WG_THREADS), lds2, stream, q);
     if (q.split_p > 1) {
       const long total = q.K * q.RSC;
       const int blocks = (int)i64min((total + 255) / 256, 2048);
-      hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
+      const int psb = (q.split_p + 63) / 64;
+      hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks, psb),
+                         dim3(256), 0,
                          stream, part_ws, dw, total, total, q.split_p);
     }
     HIP_CHECK_LAST();
