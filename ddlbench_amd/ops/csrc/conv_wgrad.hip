@@ -225,14 +225,15 @@ DEV bf16x8w wg_tr_frag(const short* img, int group, int ks, int lane) {
 // TKt x TRt dw tile, WR x WC waves. The 128x128 tile halves both
 // operands' cross-tile re-staging (the 64x64 tile is staging-BW-bound:
 // A re-reads x nr, B x nk).
-template <int TKt, int TRt, int WR, int WC>
-__global__ __launch_bounds__(WG_THREADS, 2)
+template <int TKt, int TRt, int WR, int WC, int TPB>
+__global__ __launch_bounds__(TPB, 2)
 void conv_wgrad2_kernel(WgradParams q) {
-  constexpr int GA = TKt / 32;       // A glds chunks per thread per step
-  constexpr int GB = TRt / 32;       // B
+  constexpr int GA = TKt * 8 / TPB;  // A glds chunks per thread per step
+  constexpr int GB = TRt * 8 / TPB;  // B
   constexpr int MFS = TKt / WR / 16; // A fragments per wave
   constexpr int NFS = TRt / WC / 16; // B fragments per wave
-  static_assert(WR * WC == 4, "4 waves");
+  static_assert(WR * WC == TPB / 64, "wave count");
+  static_assert(GA >= 1 && GB >= 1, "tile vs threads");
   // 3-buffer ring: [buf][A TKtx64 | B TRtx64] bf16
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* lds = reinterpret_cast<short*>(smem);
@@ -268,7 +269,7 @@ void conv_wgrad2_kernel(WgradParams q) {
   long b_pabs[GB];
 #pragma unroll
   for (int j = 0; j < GA; ++j) {
-    const int d = j * WG_THREADS + tid;
+    const int d = j * TPB + tid;
     wg_decode_chunk(d, a_p[j], a_ch[j]);
     a_chok[j] = kb + a_ch[j] < q.K;
     a_src[j] = q.dy + (p_begin + a_p[j]) * q.K + kb + a_ch[j];
@@ -276,7 +277,7 @@ void conv_wgrad2_kernel(WgradParams q) {
   }
 #pragma unroll
   for (int j = 0; j < GB; ++j) {
-    const int d = j * WG_THREADS + tid;
+    const int d = j * TPB + tid;
     int bch;
     wg_decode_chunk(d, bp_[j], bch);
     const long rsc = rb + bch;
@@ -303,7 +304,7 @@ void conv_wgrad2_kernel(WgradParams q) {
     const long plim = p_end;
 #pragma unroll
     for (int j = 0; j < GA; ++j) {
-      const int d = j * WG_THREADS + tid;
+      const int d = j * TPB + tid;
       const bf16* src =
           (a_chok[j] && a_pabs[j] + (long)t * WG_BKP < plim)
               ? a_src[j] + (long)t * WG_BKP * q.K
@@ -314,7 +315,7 @@ void conv_wgrad2_kernel(WgradParams q) {
     }
 #pragma unroll
     for (int j = 0; j < GB; ++j) {
-      const int d = j * WG_THREADS + tid;
+      const int d = j * TPB + tid;
       // advance (n,oh,ow) to step t lazily: ow' = ow + t*BP with carry
       // would need history; instead recompute from the running state —
       // the stage calls are strictly t = 0,1,2,... so the running state
@@ -468,12 +469,14 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
     split = i64max(split, 1);
     q.split_p = (int)split;
     const size_t lds2 = 3 * WG_BKP * (TKt + TRt) * sizeof(bf16);
+    // big tile: 512 threads (8 waves = 2/SIMD at the 96 KiB-LDS
+    // 1-block/CU occupancy); small tile: 256 threads x 3 blocks/CU
     if (big)
-      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 2>),
+      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512>),
                          dim3((unsigned)(nk2 * nr2 * q.split_p)),
-                         dim3(WG_THREADS), lds2, stream, q);
+                         dim3(512), lds2, stream, q);
     else
-      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4>),
+      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256>),
                          dim3((unsigned)(nk2 * nr2 * q.split_p)),
                          dim3(WG_THREADS), lds2, stream, q);
     if (q.split_p > 1) {
