@@ -186,7 +186,7 @@ DEV const bf16* b_slot_addr(const ConvParams& p, const Cursor& cu,
 // COMB_B (BN_ == 64): the B slot is too small for one glds per thread
 // per k-half, so one stage call covers BOTH k-halves (waves 0-3 fill
 // k0, waves 4-7 fill k1) and B's cursor advances by 64.
-template <int MODE, int BM_, int BN_, int WR, int WC>
+template <int MODE, int BM_, int BN_, int WR, int WC, bool PH2 = false>
 __global__ __launch_bounds__(THREADS2, 2)
 void conv_igemm2_kernel(ConvParams p) {
   constexpr int WM = BM_ / WR;       // wave tile M
@@ -309,15 +309,15 @@ void conv_igemm2_kernel(ConvParams p) {
   // ---- main loop: 4 phases per K-tile ----------------------------------
   // PH(msub, kh): ds_read fragments, issue one half-slot prefetch,
   // barrier, MFMA cluster, (counted vmcnt), barrier.
-#define PH(MSUB, KH, STAGE, VMW)                                          \
+#define PH(M0, MC, KH, STAGE, VMW)                                       \
   do {                                                                    \
     bf16* la = aslot(cbuf, KH);                                           \
-    bf16x8 af[MH];                                                        \
+    bf16x8 af[MC];                                                        \
     _Pragma("unroll")                                                     \
-    for (int mf = 0; mf < MH; ++mf)                                       \
+    for (int mf = 0; mf < (MC); ++mf)                                     \
       af[mf] = *reinterpret_cast<const bf16x8*>(                          \
-          la + (awoff + (MSUB) * (WM / 2) + mf * 16 + fr) * 32 + fchunk); \
-    if (MSUB == 0) {                                                      \
+          la + (awoff + (M0) * (WM / 2) + mf * 16 + fr) * 32 + fchunk);   \
+    if ((M0) == 0) {                                                      \
       bf16* lb = bslot(cbuf, KH);                                         \
       _Pragma("unroll")                                                   \
       for (int nf = 0; nf < NF; ++nf)                                     \
@@ -328,12 +328,12 @@ void conv_igemm2_kernel(ConvParams p) {
     __builtin_amdgcn_s_barrier();                                         \
     __builtin_amdgcn_s_setprio(1);                                        \
     _Pragma("unroll")                                                     \
-    for (int mf = 0; mf < MH; ++mf)                                       \
+    for (int mf = 0; mf < (MC); ++mf)                                     \
       _Pragma("unroll")                                                   \
       for (int nf = 0; nf < NF; ++nf)                                     \
-        acc[(MSUB) * MH + mf][nf] =                                       \
+        acc[(M0) * MH + mf][nf] =                                         \
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(                      \
-                af[mf], bfr[nf], acc[(MSUB) * MH + mf][nf], 0, 0, 0);     \
+                af[mf], bfr[nf], acc[(M0) * MH + mf][nf], 0, 0, 0);       \
     __builtin_amdgcn_s_setprio(0);                                        \
     VMW;                                                                  \
     __builtin_amdgcn_s_barrier();                                         \
@@ -348,22 +348,40 @@ void conv_igemm2_kernel(ConvParams p) {
     const int cbuf = t & 1;
     const int nxt = cbuf ^ 1;
     const bool more = (t + 1 < nsteps);
-    if (more) {
-      PH(0, 0, stage_a(nxt, 0), );
-      PH(1, 0, stage_b(nxt, 0), wait_vmcnt<V1>());
-      PH(0, 1, stage_a(nxt, 1), );
-      if (COMB_B) {
-        PH(1, 1, , wait_vmcnt<V3>());
+    if (PH2) {
+      // 2 phases per K-tile (full-M x k-half, 2x the MFMA per barrier
+      // pair): both operand halves stage inside one phase, one counted
+      // vmcnt per phase, 1-phase prefetch slack. COMB_B stages B (both
+      // halves) once, in ph0.
+      if (more && COMB_B) {
+        PH(0, MF, 0, (stage_a(nxt, 0), stage_b(nxt, 0)),
+           wait_vmcnt<GA + 1>());
+        PH(0, MF, 1, stage_a(nxt, 1), wait_vmcnt<GA>());
+      } else if (more) {
+        PH(0, MF, 0, (stage_a(nxt, 0), stage_b(nxt, 0)),
+           wait_vmcnt<V1>());
+        PH(0, MF, 1, (stage_a(nxt, 1), stage_b(nxt, 1)),
+           wait_vmcnt<V1>());
       } else {
-        PH(1, 1, stage_b(nxt, 1), wait_vmcnt<V3>());
+        PH(0, MF, 0, , wait_vmcnt<0>());
+        PH(0, MF, 1, , );
+      }
+    } else if (more) {
+      PH(0, MH, 0, stage_a(nxt, 0), );
+      PH(1, MH, 0, stage_b(nxt, 0), wait_vmcnt<V1>());
+      PH(0, MH, 1, stage_a(nxt, 1), );
+      if (COMB_B) {
+        PH(1, MH, 1, , wait_vmcnt<V3>());
+      } else {
+        PH(1, MH, 1, stage_b(nxt, 1), wait_vmcnt<V3>());
       }
     } else {
       // last K-tile: nothing left to stage; the k1 halves may still be
       // in flight, so the mid-tile wait drains fully
-      PH(0, 0, , );
-      PH(1, 0, , wait_vmcnt<0>());
-      PH(0, 1, , );
-      PH(1, 1, , );
+      PH(0, MH, 0, , );
+      PH(1, MH, 0, , wait_vmcnt<0>());
+      PH(0, MH, 1, , );
+      PH(1, MH, 1, , );
     }
   }
 #undef PH
@@ -425,12 +443,17 @@ void conv_igemm2_kernel(ConvParams p) {
 
 template <int MODE>
 bool dispatch_v2(const ConvParams& p, hipStream_t stream) {
-#define LAUNCH2(BM_, BN_, WR, WC)                                           \
+  // DDLB_CONV_PH2=0 selects the 4-phase schedule for A/B (default: the
+  // 2-phase full-M schedule — half the barriers per K-tile; the PMC
+  // profile showed 37% of wave cycles parked on barriers at 4-phase)
+  const char* phe = getenv("DDLB_CONV_PH2");
+  const bool ph2 = !(phe && phe[0] == '0');
+#define LAUNCH2(BM_, BN_, WR, WC, PH2_)                                     \
   do {                                                                      \
     const long nbm = (p.M + (BM_) - 1) / (BM_);                             \
     const long nbn = (p.Nd + (BN_) - 1) / (BN_);                            \
     const size_t lds_bytes = 4 * ((BM_) + (BN_)) * 32 * sizeof(bf16);       \
-    hipLaunchKernelGGL((conv_igemm2_kernel<MODE, BM_, BN_, WR, WC>),        \
+    hipLaunchKernelGGL((conv_igemm2_kernel<MODE, BM_, BN_, WR, WC, PH2_>),  \
                        dim3((unsigned)(nbm * nbn)), dim3(THREADS2),         \
                        lds_bytes, stream, p);                               \
   } while (0)
@@ -439,9 +462,11 @@ bool dispatch_v2(const ConvParams& p, hipStream_t stream) {
   // 256x128 tile (acc 64, ~204 VGPRs clean) serves all Nd >= 96 with
   // column blocks. Nd in [48,96) gets a 512x64 tile (combined-B).
   if (p.Nd >= 96) {
-    LAUNCH2(256, 128, 2, 4);
+    if (ph2) LAUNCH2(256, 128, 2, 4, true);
+    else LAUNCH2(256, 128, 2, 4, false);
   } else if (p.Nd >= 48) {
-    LAUNCH2(512, 64, 8, 1);
+    if (ph2) LAUNCH2(512, 64, 8, 1, true);
+    else LAUNCH2(512, 64, 8, 1, false);
   } else {
     return false;
   }
