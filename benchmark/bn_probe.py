@@ -44,8 +44,10 @@ def main():
         b = torch.randn(C, device=dev)
         rm = torch.zeros(C, device=dev)
         rv = torch.ones(C, device=dev)
-        y, mean, invstd = ext.bn_act_fwd(x, res, g, b, rm, rv, True, 0.1,
-                                         1e-5, 1, True)
+        outs = ext.bn_act_fwd(x, res, g, b, rm, rv, True, 0.1,
+                              1e-5, 1, True)
+        y, mean, invstd = outs[0], outs[1], outs[2]
+        mask = outs[3] if len(outs) > 3 else None
         nb = N * C * H * W * 2 / 1e9
         row = {"shape": f"{N}x{C}x{H}x{W}"}
         for variant in (1, 2):
@@ -55,7 +57,11 @@ def main():
             te = t_ms(lambda: ext.bn_act_fwd(x, res, g, b, rm, rv, False,
                                              0.1, 1e-5, 1, True))
             tb = t_ms(lambda: ext.bn_act_bwd(dy, y, x, mean, invstd, g, 1,
-                                             True, True, True))
+                                             True, True, True, None))
+            tbm = (t_ms(lambda: ext.bn_act_bwd(dy, None, x, mean, invstd,
+                                               g, 1, True, True, True,
+                                               mask))
+                   if mask is not None else float("nan"))
             row[f"v{variant}_fwd_ms"] = round(tf, 3)
             row[f"v{variant}_fwd_TBs"] = round(4 * nb / tf, 2)
             row[f"v{variant}_apply_ms"] = round(te, 3)
@@ -63,6 +69,7 @@ def main():
             row[f"v{variant}_stats_TBs"] = round(nb / max(tf - te, 1e-5), 2)
             row[f"v{variant}_bwd_ms"] = round(tb, 3)
             row[f"v{variant}_bwd_TBs"] = round(8 * nb / tb, 2)
+            row[f"v{variant}_bwdmask_ms"] = round(tbm, 3)
         ext.set_bn_variant(0)
         print(json.dumps(row), flush=True)
 
