@@ -955,7 +955,10 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
-  if (nhwc && sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
+  // a non-null mask FORCES the masked-capable kernel2 path (the vec
+  // fallback would read the null y)
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0 &&
+      (g_bn_variant != 1 || msk != nullptr)) {
     const int64_t rows = total / C;
     const int CG8 = (int)i64min(C / 8, 64);
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
