@@ -758,26 +758,36 @@ static inline int elementwise_grid(int64_t total, int block) {
 }
 
 
-// Row-slice block count (gridDim.y) shared by the stats/bwd-reduce
-// kernels AND the workspace allocation in the bindings: the partial
-// slab is [S][2][C] doubles. Must mirror the launchers' dispatch.
+// Reduction-kernel dispatch geometry, shared by the launchers AND the
+// workspace sizing in the bindings (partial slab = [S][2][C] doubles).
+// g_bn_variant: 0 = shape-adaptive (vec reductions for C <= 512 where
+// the probe shows them ahead), 1 = scalar, 2 = vec.
+struct BnGeom { bool vec; int64_t cblocks, S; };
+
+static BnGeom bn_reduce_geom(int64_t rows, int64_t C, int elsize) {
+  BnGeom g;
+  const bool vec_ok = (elsize == 2 && C % 8 == 0) ||
+                      (elsize == 4 && C % 4 == 0);
+  g.vec = vec_ok && (g_bn_variant == 2 ||
+                     (g_bn_variant == 0 && C <= 512 && elsize == 2));
+  if (g.vec) {
+    const int64_t lanes = elsize == 2 ? C / 8 : C / 4;
+    const int64_t CG = i64min(lanes, 64);
+    g.cblocks = (lanes + CG - 1) / CG;
+  } else {
+    const int64_t CG = C >= 64 ? 64 : C;
+    g.cblocks = (C + CG - 1) / CG;
+  }
+  // rows/16 (not /512): a block still covers >=16 rows per row-group,
+  // and the old cap starved the small-HW layers (C2048@7x7 ran 196
+  // blocks on a 256-CU chip)
+  g.S = i64min(i64max(rows / 16, 1), i64max(2048 / g.cblocks, 1));
+  return g;
+}
+
 int64_t bn_reduce_gridS(int64_t N, int64_t C, int64_t HW, int nhwc,
                         int elsize) {
-  if (nhwc) {
-    const int64_t rows = N * HW;
-    int64_t cblocks;
-    if (elsize == 2 && C % 8 == 0 && g_bn_variant == 2) {
-      const int64_t CG8 = i64min(C / 8, 64);
-      cblocks = (C / 8 + CG8 - 1) / CG8;
-    } else if (elsize == 4 && C % 4 == 0 && g_bn_variant == 2) {
-      const int64_t CG4 = i64min(C / 4, 64);
-      cblocks = (C / 4 + CG4 - 1) / CG4;
-    } else {
-      const int64_t CG = C >= 64 ? 64 : C;
-      cblocks = (C + CG - 1) / CG;
-    }
-    return i64min(i64max(rows / 512, 1), i64max(2048 / cblocks, 1));
-  }
+  if (nhwc) return bn_reduce_geom(N * HW, C, elsize).S;
   int64_t S = i64min((N * HW + 255) / 256, i64max(2048 / C, 1));
   return i64max(S, 1);
 }
@@ -788,21 +798,20 @@ void launch_bn_stats(const T* x, double* sums, int64_t N, int64_t C,
   const int block = 256;
   if (nhwc) {
     const int64_t rows = N * HW;
-    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
+    const BnGeom g = bn_reduce_geom(rows, C, (int)sizeof(T));
+    if (g.vec && sizeof(T) == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
-      const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
-      hipLaunchKernelGGL((bn_stats_nhwc_vec_kernel<T>), dim3(cblocks, S),
+      hipLaunchKernelGGL((bn_stats_nhwc_vec_kernel<T>),
+                         dim3(g.cblocks, S),
                          dim3(block), 0, stream, x, sums, rows, C, CG8);
-    } else if (sizeof(T) == 4 && C % 4 == 0 && g_bn_variant == 2) {
+    } else if (g.vec && sizeof(T) == 4) {
       const int CG4 = (int)i64min(C / 4, 64);
-      const int64_t cblocks = (C / 4 + CG4 - 1) / CG4;
-      hipLaunchKernelGGL(bn_stats_nhwc_vec_f32_kernel, dim3(cblocks, S),
+      hipLaunchKernelGGL(bn_stats_nhwc_vec_f32_kernel, dim3(g.cblocks, S),
                          dim3(block), 0, stream, (const float*)x, sums,
                          rows, C, CG4);
     } else {
       const int CG = C >= 64 ? 64 : (int)C;
-      const int64_t cblocks = (C + CG - 1) / CG;
-      hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(cblocks, S),
+      hipLaunchKernelGGL((bn_stats_nhwc_kernel<T>), dim3(g.cblocks, S),
                          dim3(block), 0, stream, x, sums, rows, C, CG);
     }
   } else {
@@ -833,11 +842,16 @@ bool launch_bn_apply(const T* x, const T* res, T* y, const float* mean,
   const int64_t cdiv = nhwc ? 1 : HW;
   const bool vec8ok = (total % 8 == 0) &&
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
-  if (nhwc && sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1) {
+  const int64_t rows_nhwc = nhwc ? total / C : 0;
+  // kernel2 starves below ~32k rows (too few blocks); grid-stride vec
+  // covers that regime — unless a mask must be written
+  if (nhwc && sizeof(T) == 2 && C % 8 == 0 && g_bn_variant != 1 &&
+      (rows_nhwc >= 32768 || (msk != nullptr && act != 0) ||
+       g_bn_variant == 2)) {
     const int64_t rows = total / C;
     const int CG8 = (int)i64min(C / 8, 64);
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
-    int64_t S = i64min(i64max(rows / 256, 1), i64max(2048 / cblocks, 1));
+    int64_t S = i64min(i64max(rows / 16, 1), i64max(2048 / cblocks, 1));
     const bool add = res != nullptr;
 #define KCASE(ACT, ADD)                                                     \
     hipLaunchKernelGGL((bn_apply_nhwc_kernel2<T, ACT, ADD>),                \
@@ -898,24 +912,23 @@ void launch_bn_bwd_reduce(const T* dy, const T* y, const T* x,
   if (nhwc) {
     const int64_t rows = N * HW;
     const bool masked = msk != nullptr && act != 0;
-    if (sizeof(T) == 2 && C % 8 == 0 && g_bn_variant == 2) {
+    const BnGeom g = bn_reduce_geom(rows, C, (int)sizeof(T));
+    if (g.vec && sizeof(T) == 2) {
       const int CG8 = (int)i64min(C / 8, 64);
-      const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
 #define CASE(ACT, MSK)                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_vec_kernel<T, ACT, MSK>),      \
-                         dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
-                         x, mean, invstd, sums, msk, rows, C, CG8)
+                         dim3(g.cblocks, S), dim3(block), 0, stream, dy,    \
+                         y, x, mean, invstd, sums, msk, rows, C, CG8)
       if (act == 0) CASE(0, false);
       else if (act == 1) { if (masked) CASE(1, true); else CASE(1, false); }
       else { if (masked) CASE(2, true); else CASE(2, false); }
 #undef CASE
     } else {
       const int CG = C >= 64 ? 64 : (int)C;
-      const int64_t cblocks = (C + CG - 1) / CG;
 #define CASE(ACT, MSK)                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_kernel<T, ACT, MSK>),          \
-                         dim3(cblocks, S), dim3(block), 0, stream, dy, y,   \
-                         x, mean, invstd, sums, msk, rows, C, CG)
+                         dim3(g.cblocks, S), dim3(block), 0, stream, dy,    \
+                         y, x, mean, invstd, sums, msk, rows, C, CG)
       if (act == 0) CASE(0, false);
       else if (act == 1) { if (masked) CASE(1, true); else CASE(1, false); }
       else { if (masked) CASE(2, true); else CASE(2, false); }
@@ -957,12 +970,14 @@ void launch_bn_bwd_dx(const T* dy, const T* y, const T* x, const float* mean,
                       (nhwc ? (C % 8 == 0) : (HW % 8 == 0));
   // a non-null mask FORCES the masked-capable kernel2 path (the vec
   // fallback would read the null y)
+  const int64_t rows_nhwc = nhwc ? total / C : 0;
   if (nhwc && sizeof(T) == 2 && C % 8 == 0 &&
-      (g_bn_variant != 1 || msk != nullptr)) {
+      ((g_bn_variant != 1 && rows_nhwc >= 32768) || msk != nullptr ||
+       g_bn_variant == 2)) {
     const int64_t rows = total / C;
     const int CG8 = (int)i64min(C / 8, 64);
     const int64_t cblocks = (C / 8 + CG8 - 1) / CG8;
-    int64_t S = i64min(i64max(rows / 256, 1), i64max(2048 / cblocks, 1));
+    int64_t S = i64min(i64max(rows / 16, 1), i64max(2048 / cblocks, 1));
     const bool add = dres != nullptr;
     const bool masked = msk != nullptr && act != 0;
 #define KCASE(ACT, ADD, MSK)                                                \
