@@ -51,9 +51,10 @@ def parse_args():
                         "pipeline across all visible GPUs")
     p.add_argument("--microbatches", type=int, default=8,
                    help="gpipe chunk count")
-    p.add_argument("--conv", default="miopen", choices=["miopen", "mfma"],
-                   help="conv backend: library (MIOpen) or the in-tree "
-                        "MFMA implicit-GEMM kernels")
+    p.add_argument("--conv", default="mfma", choices=["miopen", "mfma"],
+                   help="conv backend: the in-tree MFMA implicit-GEMM "
+                        "kernels (default — the hand-written CDNA4 hot "
+                        "path) or the MIOpen library")
     p.add_argument("--memory-format", default="channels_last",
                    choices=["channels_last", "contiguous"],
                    help="channels_last (NHWC) keeps MIOpen on its native "

@@ -1,9 +1,8 @@
 """MFMA implicit-GEMM convolution: autograd bridge + model converter.
 
-Forward and data-grad run on the hand-written NHWC bf16 kernel
-(csrc/conv_mfma.hip); the weight-grad currently goes through the library
-conv backward (aten.convolution_backward with only the weight mask) —
-a native MFMA wgrad kernel is the next kernel on the list.
+Forward and data-grad run on the hand-written NHWC bf16 kernels
+(csrc/conv_mfma*.hip) and the weight-grad on the transpose-read kernel
+(csrc/conv_wgrad.hip) — the full conv triple is in-tree by default.
 
 Eligibility: bf16, channels_last, groups=1, dilation=1, C%8==0, K%8==0.
 ``convert_convs(model)`` swaps every eligible nn.Conv2d for Conv2dMFMA
@@ -19,9 +18,9 @@ import torch.nn.functional as F
 
 from ddlbench_amd import ops as _ops
 
-# weight-grad backend: "mfma" runs the native kernel (conv_wgrad.hip),
-# anything else the library conv backward
-_WGRAD = os.environ.get("DDLB_WGRAD", "library")
+# weight-grad backend: the native kernel (conv_wgrad.hip) by default —
+# DDLB_WGRAD=library opts back into aten's conv backward
+_WGRAD = os.environ.get("DDLB_WGRAD", "mfma")
 
 _CL = torch.channels_last
 

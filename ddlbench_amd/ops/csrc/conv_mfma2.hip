@@ -186,7 +186,10 @@ DEV const bf16* b_slot_addr(const ConvParams& p, const Cursor& cu,
 // COMB_B (BN_ == 64): the B slot is too small for one glds per thread
 // per k-half, so one stage call covers BOTH k-halves (waves 0-3 fill
 // k0, waves 4-7 fill k1) and B's cursor advances by 64.
-template <int MODE, int BM_, int BN_, int WR, int WC, bool PH2 = false>
+// LIN: 1x1 stride-1 pad-0 conv — the implicit-GEMM A operand is the
+// plain row-major tensor (no im2col arithmetic, no bounds tests)
+template <int MODE, int BM_, int BN_, int WR, int WC, bool PH2 = false,
+          bool LIN = false>
 __global__ __launch_bounds__(THREADS2, 2)
 void conv_igemm2_kernel(ConvParams p) {
   constexpr int WM = BM_ / WR;       // wave tile M
@@ -247,7 +250,14 @@ void conv_igemm2_kernel(ConvParams p) {
 #pragma unroll
   for (int j = 0; j < GA; ++j) {
     const int ca = (wid * GA + j) * 64 + lane;
-    a_slot_init<MODE>(p, bm + (ca >> 2), abase[j], auv[j]);
+    if (LIN) {
+      const long m = bm + (ca >> 2);
+      const long cols = (MODE == 0) ? p.Cin : p.K;
+      abase[j] = (unsigned)((m < p.M ? m : 0) * cols);
+      auv[j] = m < p.M;
+    } else {
+      a_slot_init<MODE>(p, bm + (ca >> 2), abase[j], auv[j]);
+    }
   }
 #pragma unroll
   for (int j = 0; j < GB; ++j) {
@@ -261,9 +271,16 @@ void conv_igemm2_kernel(ConvParams p) {
   auto stage_a = [&](int buf, int kh) {
     bf16* la = aslot(buf, kh);
 #pragma unroll
-    for (int j = 0; j < GA; ++j)
-      glds16(a_slot_addr<MODE>(p, acur, abase[j], auv[j]),
-             la + ((wid * GA + j) * 64 + lane) * 8);
+    for (int j = 0; j < GA; ++j) {
+      const bf16* src;
+      if (LIN)
+        src = (auv[j] && acur.r == 0)
+                  ? p.a + abase[j] + (unsigned)acur.c
+                  : p.zero;
+      else
+        src = a_slot_addr<MODE>(p, acur, abase[j], auv[j]);
+      glds16(src, la + ((wid * GA + j) * 64 + lane) * 8);
+    }
     cursor_advance<MODE>(p, acur);
   };
   // non-combined: stages one k-half. combined: waves 0-3 fill k0 and
@@ -448,14 +465,23 @@ bool dispatch_v2(const ConvParams& p, hipStream_t stream) {
   // profile showed 37% of wave cycles parked on barriers at 4-phase)
   const char* phe = getenv("DDLB_CONV_PH2");
   const bool ph2 = !(phe && phe[0] == '0');
+  const bool lin = (MODE != 2 && p.R == 1 && p.S == 1 && p.stride == 1 &&
+                    p.pad == 0);
 #define LAUNCH2(BM_, BN_, WR, WC, PH2_)                                     \
   do {                                                                      \
     const long nbm = (p.M + (BM_) - 1) / (BM_);                             \
     const long nbn = (p.Nd + (BN_) - 1) / (BN_);                            \
     const size_t lds_bytes = 4 * ((BM_) + (BN_)) * 32 * sizeof(bf16);       \
-    hipLaunchKernelGGL((conv_igemm2_kernel<MODE, BM_, BN_, WR, WC, PH2_>),  \
-                       dim3((unsigned)(nbm * nbn)), dim3(THREADS2),         \
-                       lds_bytes, stream, p);                               \
+    if (lin)                                                                \
+      hipLaunchKernelGGL(                                                   \
+          (conv_igemm2_kernel<MODE, BM_, BN_, WR, WC, PH2_, true>),         \
+          dim3((unsigned)(nbm * nbn)), dim3(THREADS2),                      \
+          lds_bytes, stream, p);                                            \
+    else                                                                    \
+      hipLaunchKernelGGL(                                                   \
+          (conv_igemm2_kernel<MODE, BM_, BN_, WR, WC, PH2_, false>),        \
+          dim3((unsigned)(nbm * nbn)), dim3(THREADS2),                      \
+          lds_bytes, stream, p);                                            \
   } while (0)
   // BN=256 (acc 128 regs/wave) cannot fit beside the im2col staging
   // state in the 256-VGPR/2-wave budget (measured 105-reg spill); the

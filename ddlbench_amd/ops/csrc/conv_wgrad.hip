@@ -225,7 +225,9 @@ DEV bf16x8w wg_tr_frag(const short* img, int group, int ks, int lane) {
 // TKt x TRt dw tile, WR x WC waves. The 128x128 tile halves both
 // operands' cross-tile re-staging (the 64x64 tile is staging-BW-bound:
 // A re-reads x nr, B x nk).
-template <int TKt, int TRt, int WR, int WC, int TPB>
+// R1: 1x1 stride-1 pad-0 conv — B (im2col x) degenerates to the plain
+// row-major tensor; skip the per-step carry chain and bounds tests
+template <int TKt, int TRt, int WR, int WC, int TPB, bool R1 = false>
 __global__ __launch_bounds__(TPB, 2)
 void conv_wgrad2_kernel(WgradParams q) {
   constexpr int GA = TKt * 8 / TPB;  // A glds chunks per thread per step
@@ -267,6 +269,7 @@ void conv_wgrad2_kernel(WgradParams q) {
   int b_n[GB], b_oh[GB], b_ow[GB];
   bool b_rscok[GB];
   long b_pabs[GB];
+  const bf16* b_lin[GB];  // R1: linear source pointers
 #pragma unroll
   for (int j = 0; j < GA; ++j) {
     const int d = j * TPB + tid;
@@ -296,6 +299,7 @@ void conv_wgrad2_kernel(WgradParams q) {
     b_oh[j] = rem / q.OW;
     b_ow[j] = rem - b_oh[j] * q.OW;
     b_pabs[j] = pp;
+    if (R1) b_lin[j] = q.x + (p_begin + bp_[j]) * q.C + b_c0[j];
   }
 
   auto stage = [&](int buf, int t) {
@@ -321,7 +325,10 @@ void conv_wgrad2_kernel(WgradParams q) {
       // the stage calls are strictly t = 0,1,2,... so the running state
       // IS step t's state; advance after use.
       const bf16* src = q.zero;
-      if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim) {
+      if (R1) {
+        if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim)
+          src = b_lin[j] + (long)t * WG_BKP * q.C;
+      } else if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim) {
         const int ih = b_oh[j] * q.stride + b_r[j] - q.pad;
         const int iw = b_ow[j] * q.stride + b_s[j] - q.pad;
         if (ih >= 0 && ih < q.H && iw >= 0 && iw < q.W)
@@ -331,13 +338,15 @@ void conv_wgrad2_kernel(WgradParams q) {
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(lb + d * 8), 16, 0, 0);
-      // advance the pixel coords by BP rows (carry chain)
-      b_ow[j] += WG_BKP;
-      while (b_ow[j] >= q.OW) {
-        b_ow[j] -= q.OW;
-        if (++b_oh[j] == q.OH) {
-          b_oh[j] = 0;
-          ++b_n[j];
+      if (!R1) {
+        // advance the pixel coords by BP rows (carry chain)
+        b_ow[j] += WG_BKP;
+        while (b_ow[j] >= q.OW) {
+          b_ow[j] -= q.OW;
+          if (++b_oh[j] == q.OH) {
+            b_oh[j] = 0;
+            ++b_n[j];
+          }
         }
       }
     }
@@ -471,14 +480,20 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
     const size_t lds2 = 3 * WG_BKP * (TKt + TRt) * sizeof(bf16);
     // big tile: 512 threads (8 waves = 2/SIMD at the 96 KiB-LDS
     // 1-block/CU occupancy); small tile: 256 threads x 3 blocks/CU
-    if (big)
-      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512>),
-                         dim3((unsigned)(nk2 * nr2 * q.split_p)),
-                         dim3(512), lds2, stream, q);
+    const bool r1 = (R == 1 && S == 1 && stride == 1 && pad == 0);
+    const unsigned grid2 = (unsigned)(nk2 * nr2 * q.split_p);
+    if (big && r1)
+      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512, true>),
+                         dim3(grid2), dim3(512), lds2, stream, q);
+    else if (big)
+      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512, false>),
+                         dim3(grid2), dim3(512), lds2, stream, q);
+    else if (r1)
+      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256, true>),
+                         dim3(grid2), dim3(WG_THREADS), lds2, stream, q);
     else
-      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256>),
-                         dim3((unsigned)(nk2 * nr2 * q.split_p)),
-                         dim3(WG_THREADS), lds2, stream, q);
+      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256, false>),
+                         dim3(grid2), dim3(WG_THREADS), lds2, stream, q);
     if (q.split_p > 1) {
       const long total = q.K * q.RSC;
       const int blocks = (int)i64min((total + 255) / 256, 2048);
