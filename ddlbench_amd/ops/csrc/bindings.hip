@@ -101,6 +101,20 @@ T* dptr(const torch::Tensor& t) {
   return reinterpret_cast<T*>(t.data_ptr());
 }
 
+// Cached per-device 16-byte zero page for the conv kernels' padding
+// redirect (a fresh torch::zeros per call cost ~160 fill launches per
+// training step).
+torch::Tensor zero_page(const torch::Tensor& like) {
+  static std::unordered_map<int, torch::Tensor> zp;
+  const int dev = (int)like.get_device();
+  auto it = zp.find(dev);
+  if (it == zp.end()) {
+    zp[dev] = torch::zeros({16}, like.options().dtype(at::kBFloat16));
+    it = zp.find(dev);
+  }
+  return it->second;
+}
+
 // Per-device BN reduction workspace: [S][2][C] double partial slabs.
 // Reduce-style kernels write plain per-block stores and the finalize
 // kernels sum over S — no cross-block atomics (a C=64 layer at S~1500
@@ -352,7 +366,7 @@ torch::Tensor conv_igemm_fwd(torch::Tensor x, torch::Tensor w,
   auto y = torch::empty({N, K, OH, OW},
                         x.options().memory_format(
                             at::MemoryFormat::ChannelsLast));
-  auto zero = torch::zeros({16}, x.options());
+  auto zero = zero_page(x);
   launch_conv_igemm(x.data_ptr(), w.data_ptr(), y.data_ptr(),
                     zero.data_ptr(), N, H, W, C, K, OH, OW, R, S,
                     (int)stride, (int)pad, /*dgrad=*/0,
@@ -381,7 +395,7 @@ torch::Tensor conv_igemm_dgrad(torch::Tensor dy, torch::Tensor w_perm,
                          dy.options().memory_format(
                              at::MemoryFormat::ChannelsLast));
   if (stride == 2 && (R == 1 || S == 1)) dx.zero_();
-  auto zero = torch::zeros({16}, dy.options());
+  auto zero = zero_page(dy);
   launch_conv_igemm(dy.data_ptr(), w_perm.data_ptr(), dx.data_ptr(),
                     zero.data_ptr(), (int)N, (int)H, (int)W, (int)C, K,
                     OH, OW, R, S, (int)stride, (int)pad, /*dgrad=*/1,
@@ -405,8 +419,9 @@ torch::Tensor conv_igemm_wgrad(torch::Tensor x, torch::Tensor dy,
   const int K = dy.size(1), OH = dy.size(2), OW = dy.size(3);
   TORCH_CHECK(C % 8 == 0 && K % 8 == 0,
               "conv_wgrad needs C %% 8 == 0 and K %% 8 == 0");
-  auto dw = torch::zeros({K, R * S * C}, x.options().dtype(at::kFloat));
-  auto zero = torch::zeros({16}, x.options());
+  auto dw = torch::empty({K, R * S * C},
+                         x.options().dtype(at::kFloat));
+  auto zero = zero_page(x);
   // split-P partial-slab workspace (cached per device, 32 MB fp32)
   static std::unordered_map<int, torch::Tensor> wgws;
   const int devi = (int)x.get_device();

@@ -498,6 +498,10 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
       const long total = q.K * q.RSC;
       const int blocks = (int)i64min((total + 255) / 256, 2048);
       const int psb = (q.split_p + 63) / 64;
+      // multi-chunk combine accumulates with atomics -> dw must be 0
+      // (dw arrives as torch::empty; the single-chunk combine writes)
+      if (psb > 1)
+        hipMemsetAsync(dw, 0, total * sizeof(float), stream);
       hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks, psb),
                          dim3(256), 0,
                          stream, part_ws, dw, total, total, q.split_p);
@@ -509,6 +513,8 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
   long split = 2048 / i64max(nk * nr, 1);
   split = i64max(i64min(split, (q.P + WG_BKP - 1) / WG_BKP), 1);
   q.split_p = (int)split;
+  // v1 accumulates with atomics -> zero dw (arrives as torch::empty)
+  hipMemsetAsync(dw, 0, q.K * q.RSC * sizeof(float), stream);
   const size_t lds_bytes = (WG_BKP * WG_TK + WG_BKP * WG_TR) * sizeof(bf16);
   hipLaunchKernelGGL(conv_wgrad_kernel,
                      dim3((unsigned)(nk * nr * q.split_p)),
