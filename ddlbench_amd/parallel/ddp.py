@@ -32,7 +32,7 @@ import torch.nn as nn
 
 
 class _Bucket:
-    __slots__ = ("params", "flat", "views", "pending", "work")
+    __slots__ = ("params", "flat", "views", "pending", "work", "comm")
 
     def __init__(self) -> None:
         self.params: List[torch.nn.Parameter] = []
@@ -40,6 +40,7 @@ class _Bucket:
         self.views = {}
         self.pending = 0
         self.work = None
+        self.comm = None  # fp32 reduction scratch (DDLB_BUCKET_FP32=1)
 
 
 class BucketedDataParallel(nn.Module):
@@ -52,6 +53,7 @@ class BucketedDataParallel(nn.Module):
         if bucket_mb <= 0:
             bucket_mb = float(os.environ.get("DDLB_BUCKET_MB", "32"))
         self.bucket_bytes = int(bucket_mb * 2**20)
+        self._fp32_reduce = os.environ.get("DDLB_BUCKET_FP32", "0") == "1"
         self.world_size = (dist.get_world_size(self.pg)
                            if dist.is_initialized() else 1)
         self._buckets: List[_Bucket] = []
@@ -128,8 +130,18 @@ class BucketedDataParallel(nn.Module):
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0:
-            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                     group=self.pg, async_op=True)
+            # Reduction dtype: param dtype (bf16) by default — the same
+            # choice torch DDP makes, and the bandwidth-right one for
+            # the 7-link xGMI mesh. The reference's horovod reduced
+            # fp32; DDLB_BUCKET_FP32=1 reproduces that numerics at 2x
+            # the wire bytes (docs/MULTIGPU.md).
+            if self._fp32_reduce and b.flat.dtype != torch.float32:
+                b.comm = b.flat.to(torch.float32)
+                b.work = dist.all_reduce(b.comm, op=dist.ReduceOp.SUM,
+                                         group=self.pg, async_op=True)
+            else:
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.pg, async_op=True)
             b.pending = len(b.params)
 
     def finalize_backward(self) -> None:
@@ -148,7 +160,12 @@ class BucketedDataParallel(nn.Module):
             if b.work is not None:
                 b.work.wait()
                 b.work = None
-                if self.average:
+                if b.comm is not None:
+                    if self.average:
+                        b.comm.div_(self.world_size)
+                    b.flat.copy_(b.comm)
+                    b.comm = None
+                elif self.average:
                     b.flat.div_(self.world_size)
             # restore grads dropped by a zero_grad(set_to_none=True)
             for p in b.params:
