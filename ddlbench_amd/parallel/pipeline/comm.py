@@ -70,6 +70,45 @@ class PipelineTransport:
         return self.channels[(src, dst, direction)]
 
 
+class _LocalWork:
+    def wait(self):
+        return None
+
+
+class _LocalChannel:
+    def __init__(self, q):
+        self.q = q
+
+    def isend(self, t: torch.Tensor):
+        self.q.append(t.detach().clone())
+        return _LocalWork()
+
+    def irecv(self, buf: torch.Tensor):
+        buf.copy_(self.q.popleft())
+        return _LocalWork()
+
+
+class LocalTransport:
+    """In-process transport: every stage lives in ONE process. Lets a
+    single GPU run a multi-stage 1F1B schedule (schedule, weight
+    versioning and copy ordering under real HIP streams) without
+    multi-process RCCL — the sends/recvs become deque hand-offs, so the
+    caller must interleave the stages' run_forward/run_backward in a
+    valid pipeline order (a recv before its send raises IndexError
+    rather than deadlocking)."""
+
+    def __init__(self, edges: List[Tuple[int, int]]):
+        from collections import deque
+        self.channels = {}
+        for (src, dst) in edges:
+            for direction in ("fwd", "bwd"):
+                self.channels[(src, dst, direction)] = _LocalChannel(
+                    deque())
+
+    def channel(self, src: int, dst: int, direction: str):
+        return self.channels[(src, dst, direction)]
+
+
 def dry_run_shapes(stages: List[torch.nn.Module], sample: torch.Tensor,
                    device=torch.device("cpu")) -> List[torch.Size]:
     """Output shape of each stage for one micro/minibatch — run once at
