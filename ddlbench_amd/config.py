@@ -129,6 +129,10 @@ def make_parser(default_arch: str = "resnet18") -> argparse.ArgumentParser:
                         "(reference uses full-size synthetic trees; keep small "
                         "for smoke runs). -1 selects the highres variant where "
                         "supported.")
+    p.add_argument("--real-data", action="store_true",
+                   help="train from the class-per-directory tree under "
+                        "$DATADIR (reference `run.sh -s` semantics) "
+                        "instead of the synthetic stream")
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--lr-schedule", default="constant",
                    choices=["constant", "step30", "warmup"],
@@ -151,6 +155,7 @@ def config_from_args(dataset: str, strategy: str, args: argparse.Namespace,
     highres = getattr(args, "synthetic_scale", 1.0) == -1
     kw = dict(
         arch=args.arch,
+        synthetic=not getattr(args, "real_data", False),
         synthetic_scale=(1.0 if highres else max(args.synthetic_scale, 0.0) or 1.0),
         lr=args.lr,
         lr_schedule=getattr(args, "lr_schedule", "constant"),

@@ -62,11 +62,20 @@ def make_loaders(cfg: BenchConfig, world_size: int = 1, rank: int = 0,
     """(train_loader, test_loader[, samplers]) honouring DP sharding.
 
     Mirrors the reference's DistributedSampler usage for the horovod path
-    (/root/reference/benchmark/mnist/mnist_horovod.py:209-219)."""
-    train_ds = SyntheticImageDataset(cfg.dataset, train=True, seed=cfg.seed,
-                                     scale=cfg.synthetic_scale)
-    test_ds = SyntheticImageDataset(cfg.dataset, train=False, seed=cfg.seed,
-                                    scale=cfg.synthetic_scale)
+    (/root/reference/benchmark/mnist/mnist_horovod.py:209-219). Real-data
+    mode (cfg.synthetic False + DATADIR, reference `run.sh -s`) loads a
+    class-per-directory tree instead (data/real.py)."""
+    if not cfg.synthetic and cfg.data_dir:
+        from ddlbench_amd.data.real import RealImageDataset
+        train_ds = RealImageDataset(cfg.dataset, cfg.data_dir, train=True)
+        test_ds = RealImageDataset(cfg.dataset, cfg.data_dir, train=False)
+    else:
+        train_ds = SyntheticImageDataset(cfg.dataset, train=True,
+                                         seed=cfg.seed,
+                                         scale=cfg.synthetic_scale)
+        test_ds = SyntheticImageDataset(cfg.dataset, train=False,
+                                        seed=cfg.seed,
+                                        scale=cfg.synthetic_scale)
     train_sampler = test_sampler = None
     if world_size > 1:
         train_sampler = DistributedSampler(train_ds, num_replicas=world_size,

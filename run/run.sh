@@ -10,7 +10,8 @@
 #   -n nodes       node count — single-node launcher; >1 errors out
 #   -p interval    log interval in batches                    (default 25)
 #   -m model       resnet18/34/50/101/152 | vgg11/13/16/19 | mobilenetv2
-#   -s             use "real" data scale (full-size synthetic stream)
+#   -s             REAL data from $DATADIR (reference semantics:
+#                  run/run/run.sh:39-41); synthetic stream otherwise
 #   -e epochs      epochs                                      (default 3)
 #   -B batch       per-GPU batch size (dataset default if unset)
 #   -M micro       micro-batch count for pipeline paths
@@ -23,6 +24,7 @@ set -euo pipefail
 
 BENCH=mnist; FRAMEWORK=pytorch; GPUS=1; NODES=1; LOGINTER=25
 MODEL=""; SCALE="0.01"; EPOCHS=3; BATCH=""; MICRO=""; DTYPE=float32
+REALDATA=0
 
 while getopts "b:f:g:n:p:m:se:B:M:d:h" opt; do
   case $opt in
@@ -32,7 +34,7 @@ while getopts "b:f:g:n:p:m:se:B:M:d:h" opt; do
     n) NODES=$OPTARG;;
     p) LOGINTER=$OPTARG;;
     m) MODEL=$OPTARG;;
-    s) SCALE="1.0";;
+    s) REALDATA=1;;
     e) EPOCHS=$OPTARG;;
     B) BATCH=$OPTARG;;
     M) MICRO=$OPTARG;;
@@ -60,6 +62,10 @@ ROOT="$(cd "$(dirname "$0")/.." && pwd)"
 DS=$BENCH
 EXTRA=()
 if [ "$BENCH" = highres ]; then DS=imagenet; EXTRA+=(-s -1); else EXTRA+=(-s "$SCALE"); fi
+if [ "$REALDATA" = "1" ]; then
+  : "${DATADIR:?-s (real data) needs DATADIR}"
+  EXTRA+=(--real-data)
+fi
 if [ -z "$MODEL" ]; then
   case $DS in imagenet) MODEL=resnet50;; *) MODEL=resnet18;; esac
 fi

@@ -49,3 +49,57 @@ def test_synthetic_batch():
     cfg = BenchConfig(dataset="imagenet")
     x, y = synthetic_batch(cfg, batch_size=2)
     assert x.shape == (2, 3, 224, 224) and y.shape == (2,)
+
+
+def test_real_data_tree(tmp_path):
+    """Reference `-s` semantics: class-per-directory tree under DATADIR
+    (VERDICT round-1 item 5). Tiny .npy tree on disk -> loaders."""
+    import numpy as np
+    from ddlbench_amd.config import BenchConfig
+    from ddlbench_amd.data import make_loaders
+    from ddlbench_amd.data.real import RealImageDataset
+
+    rng = np.random.default_rng(0)
+    for split, n in (("train", 4), ("val", 2)):
+        for cls in ("cat", "dog"):
+            d = tmp_path / split / cls
+            d.mkdir(parents=True)
+            for i in range(n):
+                # HWC uint8, wrong spatial size (tests resize), 3ch
+                np.save(d / f"{i}.npy",
+                        rng.integers(0, 255, (20, 24, 3), dtype=np.uint8))
+
+    ds = RealImageDataset("cifar10", str(tmp_path), train=True)
+    assert len(ds) == 8 and ds.num_classes == 2
+    x, y = ds[0]
+    assert x.shape == (3, 32, 32) and x.dtype == torch.float32
+    assert y in (0, 1)
+
+    cfg = BenchConfig(dataset="cifar10", synthetic=False,
+                      data_dir=str(tmp_path), batch_size=4,
+                      num_workers=0)
+    tr, te, _ = make_loaders(cfg, pin_memory=False)
+    xb, yb = next(iter(tr))
+    assert xb.shape == (4, 3, 32, 32)
+    assert len(te.dataset) == 4  # val split
+
+
+def test_real_data_grayscale_and_pt(tmp_path):
+    import numpy as np
+    from ddlbench_amd.data.real import RealImageDataset
+    d = tmp_path / "train" / "zero"
+    d.mkdir(parents=True)
+    np.save(d / "a.npy", np.zeros((28, 28), dtype=np.uint8))
+    torch.save(torch.randn(1, 28, 28), d / "b.pt")
+    ds = RealImageDataset("mnist", str(tmp_path), train=True)
+    assert len(ds) == 2
+    for i in range(2):
+        x, y = ds[i]
+        assert x.shape == (1, 28, 28) and y == 0
+
+
+def test_real_data_missing_dir():
+    import pytest
+    from ddlbench_amd.data.real import RealImageDataset
+    with pytest.raises(FileNotFoundError):
+        RealImageDataset("mnist", "/nonexistent/datadir", train=True)
