@@ -165,3 +165,120 @@ def partition_chain(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
     pure_dp = pt[n] / M + _dp_allreduce_time(M, pp[n], bw)
     return PartitionResult(stages=stages, bottleneck=A[n][M][0],
                            pure_dp_time=pure_dp, num_gpus=M)
+
+
+def partition_dag(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
+                  memory_bytes: float = MI355X_MEM,
+                  straight: bool = False,
+                  inference: bool = False) -> PartitionResult:
+    """Partition an arbitrary (non-chain) DAG over its antichain cuts.
+
+    The reference's hierarchical optimizer enumerates antichain states
+    and runs a DP over (state, machine-count)
+    (optimizer_graph_hierarchical.py:222-332). Same idea here: every
+    cut is an antichain frontier; a stage is the difference between two
+    comparable cuts' predecessor-closures (a convex node set, so it is
+    executable contiguously); the DP minimizes the bottleneck stage
+    time including per-stage replication and the cut's activation
+    transfer."""
+    states, _adj = graph.antichain_dag()
+    all_ids = frozenset(graph.nodes.keys())
+
+    def covered(ac: frozenset) -> frozenset:
+        c = set(ac)
+        for i in ac:
+            c |= graph.predecessors(i)
+        return frozenset(c)
+
+    covs = [frozenset()] + [covered(s) for s in states]
+    if all_ids not in covs:
+        covs.append(all_ids)
+    order = sorted(range(len(covs)), key=lambda i: len(covs[i]))
+    end_idx = next(i for i in order if covs[i] == all_ids)
+
+    # bytes crossing a cut = every covered node with an edge into the
+    # uncovered side (the frontier antichain alone under-counts — the
+    # reference augments antichains for the same reason,
+    # graph.py:350-366)
+    def cut_bytes(cov: frozenset) -> float:
+        return sum(graph.nodes[i].activation_size for i in cov
+                   if any(d not in cov for d in graph.edges[i]))
+
+    cut_act_of = [cut_bytes(c) for c in covs]
+
+    t = {i: (nd.fwd_time if inference else nd.compute_time)
+         for i, nd in graph.nodes.items()}
+    par = {i: nd.parameter_size for i, nd in graph.nodes.items()}
+    act = {i: nd.activation_size for i, nd in graph.nodes.items()}
+
+    INF = float("inf")
+    M = num_gpus
+    # A[state][m] = (bottleneck, prev_state, replicas)
+    A = {si: [(INF, -1, 0)] * (M + 1) for si in range(len(covs))}
+    empty = covs.index(frozenset())
+    for m in range(M + 1):
+        A[empty][m] = (0.0, -1, 0)
+
+    for bi in order:
+        if covs[bi] == frozenset():
+            continue
+        for ai in order:
+            if ai == bi or not covs[ai] < covs[bi]:
+                continue
+            stage_nodes = covs[bi] - covs[ai]
+            T = sum(t[i] for i in stage_nodes)
+            P = sum(par[i] for i in stage_nodes)
+            AB = sum(act[i] for i in stage_nodes)
+            cut_act = cut_act_of[ai]
+            for m in range(1, M + 1):
+                for r in ((1,) if straight else range(1, m + 1)):
+                    prev = A[ai][m - r]
+                    if prev[0] == INF:
+                        continue
+                    stage_time = T / r + (
+                        0.0 if inference else
+                        _dp_allreduce_time(r, P, bw))
+                    act_xfers = 1.0 if inference else 2.0
+                    comm_in = (act_xfers * cut_act / (bw * r)
+                               if covs[ai] else 0.0)
+                    stash = 0 if inference else max(M - m + 1, 1)
+                    if (stash + 1) * (P + AB / max(r, 1)) > memory_bytes:
+                        continue
+                    cost = max(prev[0], stage_time, comm_in)
+                    if cost < A[bi][m][0] - 1e-15:
+                        A[bi][m] = (cost, ai, r)
+
+    if A[end_idx][M][0] == INF:
+        raise RuntimeError("no feasible DAG partition (memory bound?)")
+    stages_rev: List[Stage] = []
+    bi, m = end_idx, M
+    topo_pos = {nd.node_id: k
+                for k, nd in enumerate(graph.topological_sort())}
+    while covs[bi] != frozenset():
+        cost, ai, r = A[bi][m]
+        stage_nodes = sorted(covs[bi] - covs[ai],
+                             key=lambda i: topo_pos[i])
+        T = sum(t[i] for i in stage_nodes)
+        P = sum(par[i] for i in stage_nodes)
+        stages_rev.append(Stage(layers=list(stage_nodes), replicas=r,
+                                time=T / r + _dp_allreduce_time(r, P,
+                                                                bw)))
+        bi, m = ai, m - r
+    stages = list(reversed(stages_rev))
+    for si, st in enumerate(stages):
+        for l in st.layers:
+            graph.nodes[l].stage_id = si
+
+    total_t = sum(t.values())
+    pure_dp = total_t / M + _dp_allreduce_time(
+        M, sum(par.values()), bw)
+    return PartitionResult(stages=stages, bottleneck=A[end_idx][M][0],
+                           pure_dp_time=pure_dp, num_gpus=M)
+
+
+def partition_graph(graph: Graph, num_gpus: int, **kw) -> PartitionResult:
+    """Chain-or-DAG dispatch: the reference partitions arbitrary DAGs;
+    chains take the exact contiguous DP."""
+    if graph.is_chain():
+        return partition_chain(graph, num_gpus, **kw)
+    return partition_dag(graph, num_gpus, **kw)

@@ -142,3 +142,83 @@ def profile_sequential(seq: nn.Sequential, sample: torch.Tensor,
             parameter_size=float(param_bytes[i]),
         ))
     return Graph.chain(nodes)
+
+
+def profile_module_graph(model: nn.Module, *sample_args,
+                         device: Optional[torch.device] = None,
+                         iters: int = 6, warmup: int = 2) -> Graph:
+    """Traced DAG + per-node fwd/bwd timings (VERDICT round-1 item 7:
+    the reference profiles arbitrary DAGs, profiler main.py:446-528).
+
+    One hooked forward captures each leaf module's REAL inputs; each
+    node is then timed in isolation (device-synchronized wall clock,
+    fwd and bwd) exactly like profile_sequential times chain layers.
+    The result is the trace_module_graph DAG with timings merged in —
+    ready for partition_graph."""
+    device = device or (sample_args[0].device if sample_args
+                        else torch.device("cpu"))
+    model = model.to(device)
+    sample_args = tuple(a.to(device) if torch.is_tensor(a) else a
+                        for a in sample_args)
+    g = trace_module_graph(model, *sample_args)
+
+    # capture real inputs per leaf-module CALL (same order as the trace)
+    captured = []
+    hooks = []
+
+    def cap(module, inputs, kwargs, output):
+        ins = tuple(t.detach().clone() if torch.is_tensor(t) else t
+                    for t in inputs)
+        kws = {k: (v.detach().clone() if torch.is_tensor(v) else v)
+               for k, v in (kwargs or {}).items()}
+        captured.append((module, ins, kws))
+
+    for name, m in model.named_modules():
+        if len(list(m.children())) == 0 and name:
+            hooks.append(m.register_forward_hook(
+                lambda mod, i, kw, o: cap(mod, i, kw, o),
+                with_kwargs=True))
+    was_training = model.training
+    model.train()
+    try:
+        with torch.no_grad():
+            model(*sample_args)
+    finally:
+        for h in hooks:
+            h.remove()
+        model.train(was_training)
+
+    nodes = g.topological_sort()
+    assert len(captured) == len(g.nodes), (len(captured), len(g.nodes))
+    for node_id in range(len(captured)):
+        module, ins, kws = captured[node_id]
+        fwd = bwd = 0.0
+        for it in range(warmup + iters):
+            gins = tuple(t.detach().requires_grad_(
+                             t.is_floating_point())
+                         if torch.is_tensor(t) else t for t in ins)
+            gkws = {k: (v.detach().requires_grad_(v.is_floating_point())
+                        if torch.is_tensor(v) else v)
+                    for k, v in kws.items()}
+            _sync(device)
+            t0 = time.perf_counter()
+            out = module(*gins, **gkws)
+            _sync(device)
+            t1 = time.perf_counter()
+            outs = out if isinstance(out, (tuple, list)) else (out,)
+            grads = [torch.ones_like(t) for t in outs
+                     if torch.is_tensor(t) and t.requires_grad]
+            ts = [t for t in outs
+                  if torch.is_tensor(t) and t.requires_grad]
+            t2 = time.perf_counter()
+            if ts:
+                torch.autograd.backward(ts, grads)
+            _sync(device)
+            t3 = time.perf_counter()
+            if it >= warmup:
+                fwd += t1 - t0
+                bwd += t3 - t2
+        g.nodes[node_id].fwd_time = fwd / iters
+        g.nodes[node_id].bwd_time = bwd / iters
+    del nodes
+    return g

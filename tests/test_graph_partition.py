@@ -201,3 +201,52 @@ def test_to_dot():
     assert dot.startswith("digraph")
     assert "n0 -> n1;" in dot
     assert "1000.00 ms" in dot  # _chain times are in seconds
+
+
+def test_partition_dag_diamond():
+    """Antichain-state DP splits a fork/join DAG into valid convex
+    stages (reference optimizer_graph_hierarchical.py:222-332)."""
+    from ddlbench_amd.parallel.pipeline.partition import (partition_dag,
+                                                          partition_graph)
+    g = Graph()
+    for i, t in enumerate([1.0, 2.0, 2.0, 1.0]):
+        g.add_node(Node(i, fwd_time=t, bwd_time=0.0,
+                        activation_size=8.0, parameter_size=8.0))
+    g.add_edge(0, 1)
+    g.add_edge(0, 2)
+    g.add_edge(1, 3)
+    g.add_edge(2, 3)
+    r = partition_dag(g, 2, bw=1e12, memory_bytes=1e15)
+    assert len(r.stages) == 2
+    assert r.bottleneck == pytest.approx(3.0)
+    # convexity: no edge from a later stage back into an earlier one
+    for src, dsts in g.edges.items():
+        for dst in dsts:
+            assert g.nodes[src].stage_id <= g.nodes[dst].stage_id
+    # dispatch picks the DAG DP for non-chains
+    r2 = partition_graph(g, 2, bw=1e12, memory_bytes=1e15)
+    assert r2.bottleneck == pytest.approx(3.0)
+
+
+def test_partition_traced_resnet_dag():
+    """End-to-end: trace a resnet DAG, time it, partition into >= 2
+    stages (VERDICT round-1 item 7 'Done' criterion)."""
+    from ddlbench_amd.models import build_model
+    from ddlbench_amd.parallel.pipeline.partition import partition_graph
+    from ddlbench_amd.parallel.pipeline.profiler import (
+        profile_module_graph)
+    torch.manual_seed(0)
+    m = build_model("mnist", "resnet18")
+    g = profile_module_graph(m, torch.randn(2, 1, 28, 28), iters=1,
+                             warmup=0)
+    assert not g.is_chain()
+    assert all(n.fwd_time >= 0 for n in g.nodes.values())
+    assert any(n.fwd_time > 0 for n in g.nodes.values())
+    r = partition_graph(g, 2, straight=True)
+    assert len(r.stages) == 2
+    for src, dsts in g.edges.items():
+        for dst in dsts:
+            assert g.nodes[src].stage_id <= g.nodes[dst].stage_id
+    # every node assigned exactly once
+    assigned = sorted(l for s in r.stages for l in s.layers)
+    assert assigned == sorted(g.nodes.keys())
