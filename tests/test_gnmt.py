@@ -158,3 +158,50 @@ def test_gnmt_edge_specs_shapes():
             assert len(xs) == len(sp)
             for t, spec in zip(xs, sp):
                 assert tuple(t.shape) == tuple(spec.shape), (t.shape, spec)
+
+
+def test_tokenizer_roundtrip(tmp_path):
+    """Reference seq2seq/data/tokenizer.py position: specials, vocab
+    build, encode/decode, save/load."""
+    from ddlbench_amd.data.tokenizer import (BOS, EOS, PAD, Tokenizer,
+                                             UNK)
+    lines = ["the cat sat", "the dog ran", "a cat ran fast"]
+    tok = Tokenizer.build(lines)
+    ids = tok.encode("the cat flew")
+    assert ids[0] == BOS and ids[-1] == EOS
+    assert UNK in ids  # 'flew' unseen
+    assert tok.decode(ids) == "the cat <unk>"
+    p = tmp_path / "vocab.txt"
+    tok.save(str(p))
+    tok2 = Tokenizer.load(str(p))
+    assert tok2.encode("the cat flew") == ids
+    assert tok.stoi["<pad>"] == PAD
+
+
+def test_text_translation_dataset(tmp_path):
+    """Reference seq2seq/data/dataset.py position: padded parallel
+    corpus with true lengths, same item contract as the synthetic
+    stream (works with BucketingSampler)."""
+    from ddlbench_amd.data.tokenizer import PAD, TextTranslationDataset
+    from ddlbench_amd.data.translation import BucketingSampler
+    (tmp_path / "train.src").write_text(
+        "the cat sat on the mat\nhello world\na b c d e\n")
+    (tmp_path / "train.tgt").write_text(
+        "die katze sass auf der matte\nhallo welt\nf g h i j\n")
+    ds = TextTranslationDataset(str(tmp_path), "train")
+    assert len(ds) == 3
+    src, slen, tgt = ds[0]
+    assert src.dtype == torch.long and tgt.dtype == torch.long
+    assert int(slen) == 8  # 6 words + BOS + EOS
+    assert (src[int(slen):] == PAD).all()
+    # bucketing sampler consumes the same src_len contract
+    bs = BucketingSampler(ds, batch_size=2, seed=0)
+    batches = list(iter(bs))
+    assert sum(len(b) for b in batches) >= 2
+
+
+def test_text_translation_dataset_missing(tmp_path):
+    import pytest as _pytest
+    from ddlbench_amd.data.tokenizer import TextTranslationDataset
+    with _pytest.raises(FileNotFoundError):
+        TextTranslationDataset(str(tmp_path), "train")
