@@ -466,8 +466,10 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
   const bool use_v2 = !(v2e && v2e[0] == '0') && part_ws != nullptr;
 
   if (use_v2) {
-    const bool big = (K >= 128 && q.RSC >= 128);
-    const long TKt = big ? 128 : 64, TRt = big ? 128 : 64;
+    // tile by shape: square 128 where possible (least re-staging),
+    // rectangular when only one dim allows it
+    const bool bigK = K >= 128, bigR = q.RSC >= 128;
+    const long TKt = bigK ? 128 : 64, TRt = bigR ? 128 : 64;
     const long nk2 = (K + TKt - 1) / TKt;
     const long nr2 = (q.RSC + TRt - 1) / TRt;
     long split = 2048 / i64max(nk2 * nr2, 1);
@@ -482,18 +484,22 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
     // 1-block/CU occupancy); small tile: 256 threads x 3 blocks/CU
     const bool r1 = (R == 1 && S == 1 && stride == 1 && pad == 0);
     const unsigned grid2 = (unsigned)(nk2 * nr2 * q.split_p);
-    if (big && r1)
-      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512, true>),
-                         dim3(grid2), dim3(512), lds2, stream, q);
-    else if (big)
-      hipLaunchKernelGGL((conv_wgrad2_kernel<128, 128, 2, 4, 512, false>),
-                         dim3(grid2), dim3(512), lds2, stream, q);
-    else if (r1)
-      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256, true>),
-                         dim3(grid2), dim3(WG_THREADS), lds2, stream, q);
-    else
-      hipLaunchKernelGGL((conv_wgrad2_kernel<64, 64, 1, 4, 256, false>),
-                         dim3(grid2), dim3(WG_THREADS), lds2, stream, q);
+#define WG_LAUNCH(TK_, TR_, WR_, WC_, TPB_)                                 \
+    do {                                                                    \
+      if (r1)                                                               \
+        hipLaunchKernelGGL(                                                 \
+            (conv_wgrad2_kernel<TK_, TR_, WR_, WC_, TPB_, true>),           \
+            dim3(grid2), dim3(TPB_), lds2, stream, q);                      \
+      else                                                                  \
+        hipLaunchKernelGGL(                                                 \
+            (conv_wgrad2_kernel<TK_, TR_, WR_, WC_, TPB_, false>),          \
+            dim3(grid2), dim3(TPB_), lds2, stream, q);                      \
+    } while (0)
+    if (bigK && bigR) WG_LAUNCH(128, 128, 2, 4, 512);
+    else if (!bigK && bigR) WG_LAUNCH(64, 128, 1, 4, 256);
+    else if (bigK && !bigR) WG_LAUNCH(128, 64, 4, 1, 256);
+    else WG_LAUNCH(64, 64, 1, 4, 256);
+#undef WG_LAUNCH
     if (q.split_p > 1) {
       const long total = q.K * q.RSC;
       const int blocks = (int)i64min((total + 255) / 256, 2048);
