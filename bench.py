@@ -55,10 +55,12 @@ def parse_args():
                    help="conv backend: the in-tree MFMA implicit-GEMM "
                         "kernels (default — the hand-written CDNA4 hot "
                         "path) or the MIOpen library")
-    p.add_argument("--hipgraph", default="auto", choices=["auto", "off"],
+    p.add_argument("--hipgraph", default="off", choices=["auto", "off"],
                    help="capture the whole training step as a hipGraph "
-                        "when single-GPU (eager fallback on capture "
-                        "failure)")
+                        "when single-GPU (measured ~1% BEHIND eager at "
+                        "b256 — the grads-persistent capture pays an "
+                        "accumulate pass; kept for smaller per-GPU "
+                        "batches where launch gaps dominate)")
     p.add_argument("--memory-format", default="channels_last",
                    choices=["channels_last", "contiguous"],
                    help="channels_last (NHWC) keeps MIOpen on its native "
