@@ -197,11 +197,25 @@ def main():
     for i in range(args.warmup):
         step_fn(i)
     barrier_sync()
+    # per-step boundaries via device events (no host syncs inside the
+    # timed region; queried after the closing synchronize)
+    events = None
+    if device.type == "cuda":
+        events = [torch.cuda.Event(enable_timing=True)
+                  for _ in range(args.steps + 1)]
     t0 = time.perf_counter()
+    if events:
+        events[0].record()
     for i in range(args.steps):
         step_fn(args.warmup + i)
+        if events:
+            events[i + 1].record()
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    step_ms = None
+    if events:
+        step_ms = [events[i].elapsed_time(events[i + 1])
+                   for i in range(args.steps)]
 
     # MAX over ranks
     if world > 1:
@@ -221,6 +235,10 @@ def main():
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1e3,
+            **({"ms_per_step_std": round(float(torch.tensor(
+                    step_ms).std(unbiased=False)), 4),
+                "ms_per_step_max": round(max(step_ms), 3)}
+               if step_ms else {}),
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
