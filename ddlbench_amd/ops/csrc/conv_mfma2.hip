@@ -487,7 +487,14 @@ bool dispatch_v2(const ConvParams& p, hipStream_t stream) {
   // state in the 256-VGPR/2-wave budget (measured 105-reg spill); the
   // 256x128 tile (acc 64, ~204 VGPRs clean) serves all Nd >= 96 with
   // column blocks. Nd in [48,96) gets a 512x64 tile (combined-B).
-  if (p.Nd >= 96) {
+  // DDLB_CONV_TILE=s selects the small high-occupancy tile (128x128,
+  // acc 32/wave -> 2 blocks/CU) for A/B against the big tile
+  const char* te = getenv("DDLB_CONV_TILE");
+  const bool small_tile = te && te[0] == 's';
+  if (p.Nd >= 96 && small_tile) {
+    if (ph2) LAUNCH2(128, 128, 2, 4, true);
+    else LAUNCH2(128, 128, 2, 4, false);
+  } else if (p.Nd >= 96) {
     if (ph2) LAUNCH2(256, 128, 2, 4, true);
     else LAUNCH2(256, 128, 2, 4, false);
   } else if (p.Nd >= 48) {
