@@ -227,9 +227,11 @@ DEV bf16x8w wg_tr_frag(const short* img, int group, int ks, int lane) {
 // A re-reads x nr, B x nk).
 // R1: 1x1 stride-1 pad-0 conv — B (im2col x) degenerates to the plain
 // row-major tensor; skip the per-step carry chain and bounds tests
-template <int TKt, int TRt, int WR, int WC, int TPB, bool R1 = false>
-__global__ __launch_bounds__(TPB, 2)
+template <int TKt, int TRt, int WR, int WC, int TPB, bool R1 = false,
+          bool RING2 = false>
+__global__ __launch_bounds__(TPB, RING2 ? 4 : 2)
 void conv_wgrad2_kernel(WgradParams q) {
+  constexpr int NBUF = RING2 ? 2 : 3;
   constexpr int GA = TKt * 8 / TPB;  // A glds chunks per thread per step
   constexpr int GB = TRt * 8 / TPB;  // B
   constexpr int MFS = TKt / WR / 16; // A fragments per wave
@@ -266,7 +268,7 @@ void conv_wgrad2_kernel(WgradParams q) {
   bool a_chok[GA];
   long a_pabs[GA];
   int bp_[GB], b_r[GB], b_s[GB], b_c0[GB];
-  int b_n[GB], b_oh[GB], b_ow[GB];
+  int b_n[GB], b_ohw[GB];  // (oh << 16) | ow
   bool b_rscok[GB];
   long b_pabs[GB];
   const bf16* b_lin[GB];  // R1: linear source pointers
@@ -296,8 +298,8 @@ void conv_wgrad2_kernel(WgradParams q) {
     const long ppc = pp < q.P ? pp : 0;
     b_n[j] = (int)(ppc / ohw);
     const int rem = (int)(ppc - (long)b_n[j] * ohw);
-    b_oh[j] = rem / q.OW;
-    b_ow[j] = rem - b_oh[j] * q.OW;
+    const int oh0 = rem / q.OW;
+    b_ohw[j] = (oh0 << 16) | (rem - oh0 * q.OW);
     b_pabs[j] = pp;
     if (R1) b_lin[j] = q.x + (p_begin + bp_[j]) * q.C + b_c0[j];
   }
@@ -329,8 +331,8 @@ void conv_wgrad2_kernel(WgradParams q) {
         if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim)
           src = b_lin[j] + (long)t * WG_BKP * q.C;
       } else if (b_rscok[j] && b_pabs[j] + (long)t * WG_BKP < plim) {
-        const int ih = b_oh[j] * q.stride + b_r[j] - q.pad;
-        const int iw = b_ow[j] * q.stride + b_s[j] - q.pad;
+        const int ih = (b_ohw[j] >> 16) * q.stride + b_r[j] - q.pad;
+        const int iw = (b_ohw[j] & 0xffff) * q.stride + b_s[j] - q.pad;
         if (ih >= 0 && ih < q.H && iw >= 0 && iw < q.W)
           src = q.x + (((long)b_n[j] * q.H + ih) * q.W + iw) * q.C +
                 b_c0[j];
@@ -340,14 +342,15 @@ void conv_wgrad2_kernel(WgradParams q) {
           (__attribute__((address_space(3))) void*)(lb + d * 8), 16, 0, 0);
       if (!R1) {
         // advance the pixel coords by BP rows (carry chain)
-        b_ow[j] += WG_BKP;
-        while (b_ow[j] >= q.OW) {
-          b_ow[j] -= q.OW;
-          if (++b_oh[j] == q.OH) {
-            b_oh[j] = 0;
+        int oh = b_ohw[j] >> 16, ow = (b_ohw[j] & 0xffff) + WG_BKP;
+        while (ow >= q.OW) {
+          ow -= q.OW;
+          if (++oh == q.OH) {
+            oh = 0;
             ++b_n[j];
           }
         }
+        b_ohw[j] = (oh << 16) | ow;
       }
     }
   };
@@ -362,7 +365,7 @@ void conv_wgrad2_kernel(WgradParams q) {
 
   // ---- prologue: stage steps 0 and 1 -----------------------------------
   stage(0, 0);
-  if (nsteps > 1) {
+  if (!RING2 && nsteps > 1) {
     stage(1, 1);
     wait_step_vmcnt<GA + GB>();  // step 0 landed, step 1 in flight
   } else {
@@ -371,8 +374,8 @@ void conv_wgrad2_kernel(WgradParams q) {
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < nsteps; ++t) {
-    const short* la = aimg(t % 3);
-    const short* lb = bimg(t % 3);
+    const short* la = aimg(t % NBUF);
+    const short* lb = bimg(t % NBUF);
     // fragments via hardware transpose read; process the two 32-p
     // halves (ks) back to back
 #pragma unroll
@@ -395,7 +398,13 @@ void conv_wgrad2_kernel(WgradParams q) {
               af[mf], bf_[nf], acc[mf][nf], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    if (t + 2 < nsteps) {
+    if (RING2) {
+      // 2-buffer ring: stage t+1 after this step's MFMAs, drain, swap.
+      // Overlap comes from the co-resident second block, not from
+      // loads spanning barriers.
+      if (t + 1 < nsteps) stage((t + 1) % 2, t + 1);
+      wait_step_vmcnt<0>();
+    } else if (t + 2 < nsteps) {
       stage((t + 2) % 3, t + 2);
       wait_step_vmcnt<GA + GB>();
     } else {
@@ -479,7 +488,9 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
       split = i64min(split, part_cap / i64max(q.K * q.RSC, 1));
     split = i64max(split, 1);
     q.split_p = (int)split;
-    const size_t lds2 = 3 * WG_BKP * (TKt + TRt) * sizeof(bf16);
+    const bool bigsq = bigK && bigR;
+    const size_t lds2 =
+        (bigsq ? 2 : 3) * WG_BKP * (TKt + TRt) * sizeof(bf16);
     // big tile: 512 threads (8 waves = 2/SIMD at the 96 KiB-LDS
     // 1-block/CU occupancy); small tile: 256 threads x 3 blocks/CU
     const bool r1 = (R == 1 && S == 1 && stride == 1 && pad == 0);
@@ -495,7 +506,16 @@ void launch_conv_wgrad(const void* x, const void* dy, float* dw,
             (conv_wgrad2_kernel<TK_, TR_, WR_, WC_, TPB_, false>),          \
             dim3(grid2), dim3(TPB_), lds2, stream, q);                      \
     } while (0)
-    if (bigK && bigR) WG_LAUNCH(128, 128, 2, 4, 512);
+    if (bigsq) {
+      if (r1)
+        hipLaunchKernelGGL(
+            (conv_wgrad2_kernel<128, 128, 2, 4, 512, true, true>),
+            dim3(grid2), dim3(512), lds2, stream, q);
+      else
+        hipLaunchKernelGGL(
+            (conv_wgrad2_kernel<128, 128, 2, 4, 512, false, true>),
+            dim3(grid2), dim3(512), lds2, stream, q);
+    }
     else if (!bigK && bigR) WG_LAUNCH(64, 128, 1, 4, 256);
     else if (bigK && !bigR) WG_LAUNCH(128, 64, 4, 1, 256);
     else WG_LAUNCH(64, 64, 1, 4, 256);
