@@ -227,3 +227,109 @@ class Graph:
     def load(cls, path: str) -> "Graph":
         with open(path) as f:
             return cls.loads(f.read())
+
+
+def compress_branches(g: "Graph"):
+    """Collapse fork..join regions into super-nodes.
+
+    The reference compresses Inception-style parallel branches so the
+    antichain enumeration stays tractable
+    (/root/reference/pipedream-fork/graph/graph.py:139-227,
+    optimizer/scripts/compress_graph_branches.py). Same purpose here:
+    for every fork node f whose join j is the unique first common
+    descendant of all branches, replace the region (f..j exclusive of
+    f, inclusive of j) with ONE node carrying the summed compute/param
+    sizes and j's activation size. Applied repeatedly until no region
+    is found. Returns (compressed_graph, mapping super_node_id ->
+    [original node ids])."""
+    mapping = {i: [i] for i in g.nodes}
+    cur = g
+    while True:
+        region = _find_branch_region(cur)
+        if region is None:
+            return cur, mapping
+        f, j, members = region
+        new = Graph()
+        # keep original ids; the super node takes j's id
+        super_members: List[int] = sorted(members | {j})
+        for i, n in cur.nodes.items():
+            if i in members:
+                continue
+            if i == j:
+                new.add_node(Node(
+                    j, desc=f"super[{len(super_members)}]",
+                    fwd_time=sum(cur.nodes[m].fwd_time
+                                 for m in super_members),
+                    bwd_time=sum(cur.nodes[m].bwd_time
+                                 for m in super_members),
+                    activation_size=cur.nodes[j].activation_size,
+                    parameter_size=sum(cur.nodes[m].parameter_size
+                                       for m in super_members)))
+            else:
+                new.add_node(Node(i, desc=n.desc, fwd_time=n.fwd_time,
+                                  bwd_time=n.bwd_time,
+                                  activation_size=n.activation_size,
+                                  parameter_size=n.parameter_size))
+        for src, dsts in cur.edges.items():
+            for dst in dsts:
+                s = j if src in members else src
+                d = j if dst in members else dst
+                if s == d or (s in members) or (d in members):
+                    continue
+                if d not in new.edges[s]:
+                    new.add_edge(s, d)
+        # merge member mappings into the super node's entry
+        merged = []
+        for m in super_members:
+            merged.extend(mapping.pop(m, [m]))
+        mapping[j] = merged
+        cur = new
+
+
+def _find_branch_region(g: "Graph"):
+    """First (fork f, join j, interior members) region where every path
+    from f re-converges at j and the interior has no edges leaving the
+    region. Interior = nodes strictly between f and j."""
+    for f in (n.node_id for n in g.topological_sort()):
+        outs = g.edges[f]
+        if len(outs) < 2:
+            continue
+        # candidate joins: common successors (incl. direct) of all outs
+        succ_sets = []
+        for o in outs:
+            s = g.successors(o) | {o}
+            succ_sets.append(s)
+        common = set.intersection(*succ_sets)
+        if not common:
+            continue
+        # earliest common node in topo order
+        topo = [n.node_id for n in g.topological_sort()]
+        pos = {i: k for k, i in enumerate(topo)}
+        j = min(common, key=lambda i: pos[i])
+        if j == f:
+            continue
+        # interior = nodes on paths f -> j (descendants of f that are
+        # ancestors of j), excluding f and j
+        desc_f = g.successors(f)
+        anc_j = g.predecessors(j)
+        members = (desc_f & anc_j) - {f, j}
+        if not members:
+            continue
+        # validity: no interior node has an edge to outside (other than
+        # toward j/members), and nothing outside (other than f) feeds
+        # the interior
+        ok = True
+        for m in members:
+            if any(d not in members and d != j for d in g.edges[m]):
+                ok = False
+                break
+            if any(s not in members and s != f for s in g.in_edges[m]):
+                ok = False
+                break
+        # the join must be fed only from the interior or f
+        if ok and any(s not in members and s != f
+                      for s in g.in_edges[j]):
+            ok = False
+        if ok:
+            return f, j, members
+    return None

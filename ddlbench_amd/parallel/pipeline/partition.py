@@ -276,9 +276,36 @@ def partition_dag(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
                            pure_dp_time=pure_dp, num_gpus=M)
 
 
-def partition_graph(graph: Graph, num_gpus: int, **kw) -> PartitionResult:
-    """Chain-or-DAG dispatch: the reference partitions arbitrary DAGs;
-    chains take the exact contiguous DP."""
+def partition_graph(graph: Graph, num_gpus: int,
+                    compress: bool = True, **kw) -> PartitionResult:
+    """Chain-or-DAG dispatch with branch compression.
+
+    The reference compresses Inception-style fork/join regions into
+    super-nodes before partitioning (graph.py:139-227,
+    compress_graph_branches.py) so the antichain enumeration stays
+    tractable; stage ids are propagated back to the original nodes."""
     if graph.is_chain():
         return partition_chain(graph, num_gpus, **kw)
+    if compress:
+        from ddlbench_amd.parallel.pipeline.graph import compress_branches
+        cg, mapping = compress_branches(graph)
+        if len(cg.nodes) < len(graph.nodes):
+            res = (partition_chain if cg.is_chain() else partition_dag)(
+                cg, num_gpus, **kw)
+            # expand super-node stages back onto the original graph
+            stages = []
+            for st in res.stages:
+                layers = []
+                for l in st.layers:
+                    layers.extend(mapping.get(l, [l]))
+                layers.sort()
+                stages.append(Stage(layers=layers, replicas=st.replicas,
+                                    time=st.time))
+                for l in layers:
+                    graph.nodes[l].stage_id = cg.nodes[
+                        st.layers[0]].stage_id if False else                         len(stages) - 1
+            return PartitionResult(stages=stages,
+                                   bottleneck=res.bottleneck,
+                                   pure_dp_time=res.pure_dp_time,
+                                   num_gpus=res.num_gpus)
     return partition_dag(graph, num_gpus, **kw)

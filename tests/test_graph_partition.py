@@ -250,3 +250,32 @@ def test_partition_traced_resnet_dag():
     # every node assigned exactly once
     assigned = sorted(l for s in r.stages for l in s.layers)
     assert assigned == sorted(g.nodes.keys())
+
+
+def test_compress_branches_inception():
+    """Fork/join regions collapse to super-nodes (reference
+    compress_branches, graph.py:139-227) and partition_graph expands
+    stage ids back to the original nodes."""
+    from ddlbench_amd.parallel.pipeline.graph import compress_branches
+    from ddlbench_amd.parallel.pipeline.partition import partition_graph
+    g = Graph()
+    # two inception blocks in series: 0 ->{1,2,3-4}-> 5 ->{6,7}-> 8 -> 9
+    for i in range(10):
+        g.add_node(Node(i, fwd_time=1.0, parameter_size=10.0,
+                        activation_size=4.0))
+    for e in [(0, 1), (0, 2), (0, 3), (3, 4), (1, 5), (2, 5), (4, 5),
+              (5, 6), (5, 7), (6, 8), (7, 8), (8, 9)]:
+        g.add_edge(*e)
+    cg, mp = compress_branches(g)
+    assert cg.is_chain()
+    assert len(cg.nodes) == 4  # 0, super(1-5), super(6-8), 9
+    assert sorted(mp[5]) == [1, 2, 3, 4, 5]
+    r = partition_graph(g, 2, straight=True, bw=1e12,
+                        memory_bytes=1e15)
+    assert len(r.stages) == 2
+    assigned = sorted(l for s in r.stages for l in s.layers)
+    assert assigned == list(range(10))
+    # convexity on the ORIGINAL graph
+    for src, dsts in g.edges.items():
+        for dst in dsts:
+            assert g.nodes[src].stage_id <= g.nodes[dst].stage_id
