@@ -80,9 +80,9 @@ class Tokenizer:
 
 
 class TextTranslationDataset(Dataset):
-    """Padded parallel corpus with per-sample true lengths — the same
-    item contract as SyntheticTranslationDataset (src[T], src_len,
-    tgt[T]) so the GNMT runner/bucketing sampler work unchanged."""
+    """Parallel corpus with the SyntheticTranslationDataset item
+    contract — unpadded (src_ids, tgt_ids) per item plus src_len(idx)
+    for the bucketing sampler; collate_translation pads per batch."""
 
     def __init__(self, root: str, prefix: str = "train",
                  tokenizer_src: Optional[Tokenizer] = None,
@@ -123,8 +123,5 @@ class TextTranslationDataset(Dataset):
 
     def __getitem__(self, idx: int):
         s, t = self.pairs[idx]
-        src = torch.full((self.src_max,), PAD, dtype=torch.long)
-        tgt = torch.full((self.tgt_max,), PAD, dtype=torch.long)
-        src[:len(s)] = torch.tensor(s, dtype=torch.long)
-        tgt[:len(t)] = torch.tensor(t, dtype=torch.long)
-        return src, torch.tensor(len(s), dtype=torch.long), tgt
+        return (torch.tensor(s, dtype=torch.long),
+                torch.tensor(t, dtype=torch.long))

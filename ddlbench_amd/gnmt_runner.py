@@ -24,7 +24,8 @@ from ddlbench_amd.utils import AverageMeter, BenchLogger, gpu_memory_gb
 def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
              hidden=1024, layers=4, lr=0.25e-3, dtype="float32",
              device="auto", log_interval=25, seed=42, max_len=50,
-             ddp=False, kernel_backend="auto", bleu_batches=2) -> dict:
+             ddp=False, kernel_backend="auto", bleu_batches=2,
+             data_dir="") -> dict:
     from ddlbench_amd.parallel import (BucketedDataParallel,
                                        allreduce_mean_scalar,
                                        init_distributed)
@@ -41,6 +42,27 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
     dev = resolve_device(cfg, env.local_rank if env else 0)
     dt = torch.bfloat16 if dtype == "bfloat16" else torch.float32
 
+    if data_dir:
+        # real parallel corpus (reference seq2seq/data position):
+        # <data_dir>/{train,val}.{src,tgt}; vocab from the train split
+        from ddlbench_amd.data.tokenizer import TextTranslationDataset
+        train_ds = TextTranslationDataset(data_dir, "train",
+                                          max_len=max_len)
+        try:
+            val_ds = TextTranslationDataset(
+                data_dir, "val", tokenizer_src=train_ds.tok_src,
+                tokenizer_tgt=train_ds.tok_tgt, max_len=max_len)
+        except FileNotFoundError:
+            val_ds = train_ds
+        vocab = max(len(train_ds.tok_src), len(train_ds.tok_tgt), 8)
+    else:
+        train_ds = SyntheticTranslationDataset(dataset_size, vocab,
+                                               max_len=max_len,
+                                               seed=seed)
+        val_ds = SyntheticTranslationDataset(max(dataset_size // 10, 8),
+                                             vocab, max_len=max_len,
+                                             seed=seed + 1)
+
     model = GNMT(vocab_size=vocab, hidden_size=hidden,
                  num_layers=layers).to(dev)
     if dt != torch.float32:
@@ -52,11 +74,6 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
                    backend=kernel_backend)
     loss_fn = LabelSmoothingLoss(0.1)
 
-    train_ds = SyntheticTranslationDataset(dataset_size, vocab,
-                                           max_len=max_len, seed=seed)
-    val_ds = SyntheticTranslationDataset(max(dataset_size // 10, 8),
-                                         vocab, max_len=max_len,
-                                         seed=seed + 1)
     sampler = BucketingSampler(train_ds, batch_size, world, rank, seed)
     log = BenchLogger(rank)
 

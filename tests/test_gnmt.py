@@ -190,10 +190,10 @@ def test_text_translation_dataset(tmp_path):
         "die katze sass auf der matte\nhallo welt\nf g h i j\n")
     ds = TextTranslationDataset(str(tmp_path), "train")
     assert len(ds) == 3
-    src, slen, tgt = ds[0]
+    src, tgt = ds[0]
     assert src.dtype == torch.long and tgt.dtype == torch.long
-    assert int(slen) == 8  # 6 words + BOS + EOS
-    assert (src[int(slen):] == PAD).all()
+    assert ds.src_len(0) == 8  # 6 words + BOS + EOS
+    assert len(src) == 8 and (src != PAD).all()
     # bucketing sampler consumes the same src_len contract
     bs = BucketingSampler(ds, batch_size=2, seed=0)
     batches = list(iter(bs))
@@ -205,3 +205,18 @@ def test_text_translation_dataset_missing(tmp_path):
     from ddlbench_amd.data.tokenizer import TextTranslationDataset
     with _pytest.raises(FileNotFoundError):
         TextTranslationDataset(str(tmp_path), "train")
+
+
+def test_run_gnmt_real_corpus(tmp_path):
+    """run_gnmt trains from an on-disk parallel corpus (data_dir)."""
+    from ddlbench_amd.gnmt_runner import run_gnmt
+    src_lines = ["a b c", "b c d e", "c d", "a a b b", "e d c b a",
+                 "a c e", "b d", "c c c c"]
+    tgt_lines = ["x y", "y z w", "z x", "x x y", "w z y x",
+                 "x z w", "y w", "z z z"]
+    (tmp_path / "train.src").write_text("\n".join(src_lines) + "\n")
+    (tmp_path / "train.tgt").write_text("\n".join(tgt_lines) + "\n")
+    res = run_gnmt(epochs=1, batch_size=4, hidden=16, layers=2,
+                   device="cpu", log_interval=0, bleu_batches=1,
+                   data_dir=str(tmp_path))
+    assert res["samples_per_sec"] > 0
