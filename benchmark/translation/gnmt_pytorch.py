@@ -19,13 +19,17 @@ def main(ddp=False):
     p.add_argument("--dtype", default="float32", choices=["float32", "bfloat16"])
     p.add_argument("--device", default="auto")
     p.add_argument("--kernel-backend", default="auto")
+    p.add_argument("--data-dir", default=os.environ.get("DATADIR", ""),
+                   help="parallel corpus root (train.src/train.tgt); "
+                        "synthetic token stream when empty")
     a = p.parse_args()
     run_gnmt(epochs=int(os.environ.get("EPOCHS", 3)),
              batch_size=int(os.environ.get("BATCH_SIZE", 64)),
              log_interval=int(os.environ.get("LOGINTER", 25)),
              dataset_size=a.dataset_size, vocab=a.vocab, hidden=a.hidden,
              layers=a.layers, lr=a.lr, dtype=a.dtype, device=a.device,
-             max_len=a.max_len, ddp=ddp, kernel_backend=a.kernel_backend)
+             max_len=a.max_len, ddp=ddp, kernel_backend=a.kernel_backend,
+             data_dir=a.data_dir)
 
 if __name__ == "__main__":
     main(ddp=False)
