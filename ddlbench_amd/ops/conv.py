@@ -106,12 +106,16 @@ class Conv2dMFMA(nn.Module):
 
 
 def convert_convs(model: nn.Module, dtype=torch.bfloat16) -> int:
-    """Replace eligible nn.Conv2d children with Conv2dMFMA. Returns the
-    number of conversions."""
+    """Replace eligible nn.Conv2d children with Conv2dMFMA and eligible
+    nn.MaxPool2d with the native NHWC pool. Returns the number of
+    conversions."""
     count = 0
     for parent in model.modules():
         for name, child in list(parent.named_children()):
             if isinstance(child, nn.Conv2d) and mfma_eligible(child, dtype):
                 setattr(parent, name, Conv2dMFMA(child))
                 count += 1
+    if dtype == torch.bfloat16:
+        from ddlbench_amd.ops.pool import convert_maxpools
+        count += convert_maxpools(model)
     return count

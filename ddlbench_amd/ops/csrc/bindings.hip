@@ -63,6 +63,13 @@ void launch_conv_wgrad(const void*, const void*, float*, float*, long,
                        int, int, int, int, int, int, int, int, int, int,
                        hipStream_t);
 
+void launch_maxpool_fwd(const void*, void*, unsigned char*, int64_t,
+                        int64_t, int, int, int, int, int, int, int,
+                        hipStream_t);
+void launch_maxpool_bwd(const void*, const unsigned char*, void*,
+                        int64_t, int64_t, int, int, int, int, int, int,
+                        int, hipStream_t);
+
 template <typename T>
 void launch_revert_varlen(const T*, T*, const int64_t*, int64_t, int64_t,
                           int64_t, hipStream_t);
@@ -440,6 +447,47 @@ torch::Tensor conv_igemm_wgrad(torch::Tensor x, torch::Tensor dy,
   return dw;
 }
 
+// ---- NHWC max-pool ----------------------------------------------------
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k,
+                                       int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "maxpool: bf16 HIP tensors only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "maxpool: x must be channels_last");
+  const int64_t N = x.size(0), C = x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(C % 8 == 0, "maxpool needs C %% 8 == 0");
+  const int OH = (H + 2 * (int)pad - (int)k) / (int)stride + 1;
+  const int OW = (W + 2 * (int)pad - (int)k) / (int)stride + 1;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(
+                            at::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty({N * OH * OW * C},
+                          x.options().dtype(at::kByte));
+  launch_maxpool_fwd(x.data_ptr(), y.data_ptr(),
+                     idx.data_ptr<unsigned char>(), N, C, H, W, OH, OW,
+                     (int)k, (int)stride, (int)pad, cur_stream());
+  return {y, idx};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx,
+                          int64_t H, int64_t W, int64_t k,
+                          int64_t stride, int64_t pad) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16,
+              "maxpool: bf16 HIP tensors only");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "maxpool: dy must be channels_last");
+  const int64_t N = dy.size(0), C = dy.size(1);
+  const int OH = (int)dy.size(2), OW = (int)dy.size(3);
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(
+                             at::MemoryFormat::ChannelsLast));
+  launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr<unsigned char>(),
+                     dx.data_ptr(), N, C, (int)H, (int)W, OH, OW,
+                     (int)k, (int)stride, (int)pad, cur_stream());
+  return dx;
+}
+
 // ---- depthwise 3x3 ----------------------------------------------------
 torch::Tensor dw3x3_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                         bool nhwc) {
@@ -532,6 +580,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_bn_variant", &set_bn_variant,
         "BN kernel variant for A/B probes (0 auto, 1 scalar-group, 2 vec)");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act(+res) backward");
+  m.def("maxpool_fwd", &maxpool_fwd, "NHWC bf16 max-pool forward");
+  m.def("maxpool_bwd", &maxpool_bwd, "NHWC bf16 max-pool backward");
   m.def("conv_igemm_fwd", &conv_igemm_fwd,
         "NHWC bf16 MFMA implicit-GEMM conv forward");
   m.def("conv_igemm_dgrad", &conv_igemm_dgrad,

@@ -31,6 +31,7 @@ KERNEL_SOURCES = [
     "cross_entropy.hip",
     "depthwise_conv.hip",
     "fused_sgd.hip",
+    "maxpool.hip",
     "seq_utils.hip",
 ]
 sources = [str(CSRC / s) for s in KERNEL_SOURCES]
