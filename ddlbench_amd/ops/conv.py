@@ -115,7 +115,8 @@ def convert_convs(model: nn.Module, dtype=torch.bfloat16) -> int:
             if isinstance(child, nn.Conv2d) and mfma_eligible(child, dtype):
                 setattr(parent, name, Conv2dMFMA(child))
                 count += 1
-    if dtype == torch.bfloat16:
+    if dtype == torch.bfloat16 and \
+            os.environ.get("DDLB_MAXPOOL", "1") != "0":
         from ddlbench_amd.ops.pool import convert_maxpools
         count += convert_maxpools(model)
     return count

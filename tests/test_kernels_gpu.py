@@ -473,12 +473,10 @@ def test_maxpool_nhwc(shape, kgeom):
     # forward max over bf16 values is exact
     torch.testing.assert_close(y1.float().contiguous(), y2, rtol=0,
                                atol=0)
-    # bf16-rounded ties can pick a different argmax than the fp32 ref;
-    # compare gradients only where the window has a unique bf16 max
-    torch.testing.assert_close(x1.grad.float().contiguous().sum(),
-                               x2.grad.sum(), rtol=1e-2, atol=1e-1)
-    mism = (x1.grad.float().contiguous() != x2.grad).float().mean()
-    assert mism < 0.02, f"grad mismatch fraction {mism}"
+    # dx sums overlapping windows' dy in bf16 vs the reference's fp32
+    # accumulate — rounding-level tolerance
+    torch.testing.assert_close(x1.grad.float().contiguous(), x2.grad,
+                               rtol=2e-2, atol=5e-2)
 
 
 def test_maxpool_module_swap():
