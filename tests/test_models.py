@@ -131,3 +131,17 @@ def test_nasnetamobile():
     x = torch.randn(2, 3, 32, 32)
     with torch.no_grad():
         torch.testing.assert_close(m(x), seq(x), rtol=1e-5, atol=1e-5)
+
+
+def test_maxpool_module_cpu_fallback():
+    """MaxPool2dNHWC falls back to F.max_pool2d off-GPU and the
+    conversion helper swaps the stem pool."""
+    import torch
+    from ddlbench_amd.ops.pool import MaxPool2dNHWC, convert_maxpools
+    m = torch.nn.Sequential(torch.nn.MaxPool2d(3, stride=2, padding=1))
+    assert convert_maxpools(m) == 1
+    assert isinstance(m[0], MaxPool2dNHWC)
+    x = torch.randn(2, 8, 16, 16)
+    y = m(x)
+    ref = torch.nn.functional.max_pool2d(x, 3, 2, 1)
+    torch.testing.assert_close(y, ref)
