@@ -188,7 +188,6 @@ def profile_module_graph(model: nn.Module, *sample_args,
             h.remove()
         model.train(was_training)
 
-    nodes = g.topological_sort()
     assert len(captured) == len(g.nodes), (len(captured), len(g.nodes))
     for node_id in range(len(captured)):
         module, ins, kws = captured[node_id]
@@ -220,5 +219,4 @@ def profile_module_graph(model: nn.Module, *sample_args,
                 bwd += t3 - t2
         g.nodes[node_id].fwd_time = fwd / iters
         g.nodes[node_id].bwd_time = bwd / iters
-    del nodes
     return g
