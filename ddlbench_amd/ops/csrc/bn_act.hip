@@ -228,7 +228,9 @@ __global__ void bn_stats_nhwc_vec_f32_kernel(const float* __restrict__ x,
 // sums = per-(row-slice) partial slabs [S][2][C] (the reduce kernels
 // write plain stores — a C=64 layer at S~1500 slices would otherwise
 // serialize ~1500 f64 atomics per channel address)
-// one block per channel; 256 threads block-reduce the S slabs
+// one block per channel; 1024 threads block-reduce the S slabs (the
+// C-bounded grid is the parallelism bottleneck at C=64/S~2048 — a
+// 256-thread block left the chip at 1 wave/CU reading 2 MB)
 __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
@@ -236,7 +238,7 @@ __global__ void bn_finalize_kernel(const double* __restrict__ sums,
                                    float* __restrict__ running_var,
                                    int64_t C, int64_t S, double count,
                                    float eps, float momentum) {
-  __shared__ double tmp[8];
+  __shared__ double tmp[16];
   const int64_t c = blockIdx.x;
   double s = 0.0, ss = 0.0;
   for (int64_t b = threadIdx.x; b < S; b += blockDim.x) {
@@ -711,7 +713,7 @@ __global__ void bn_bwd_finalize_kernel(const double* __restrict__ sums,
                                        float* __restrict__ k,  // [3][C]
                                        int64_t C, int64_t S, double count,
                                        int training) {
-  __shared__ double tmp[8];
+  __shared__ double tmp[16];
   const int64_t c = blockIdx.x;
   double sdy = 0.0, sdyx = 0.0;
   for (int64_t b = threadIdx.x; b < S; b += blockDim.x) {
@@ -825,7 +827,7 @@ void launch_bn_finalize(double* sums, float* mean, float* invstd,
                         float* rm, float* rv, int64_t C, int64_t S,
                         double count,
                         float eps, float momentum, hipStream_t stream) {
-  const int block = 256;
+  const int block = 1024;
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((unsigned)C),
                      dim3(block), 0, stream, sums, mean, invstd, rm, rv, C,
                      S, count, eps, momentum);
@@ -951,7 +953,7 @@ void launch_bn_bwd_finalize(double* sums, const float* gamma,
                             const float* invstd, float* dgamma, float* dbeta,
                             float* k, int64_t C, int64_t S, double count,
                             int training, hipStream_t stream) {
-  const int block = 256;
+  const int block = 1024;
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((unsigned)C),
                      dim3(block), 0, stream, sums, gamma, invstd, dgamma,
                      dbeta, k, C, S, count, training);
