@@ -101,3 +101,28 @@ def collate_translation(batch):
         src[:len(s), b] = s
         tgt[:len(t), b] = t
     return src, src_len, tgt[:-1], tgt[1:]
+
+
+class StaticDistributedSampler(Sampler):
+    """Deterministic contiguous rank shards, no shuffle — the
+    reference's eval-time sampler (seq2seq/data/sampler.py
+    StaticDistributedSampler): rank r takes indices [r*per, (r+1)*per)
+    of the dataset padded up to world*ceil(n/world), padding indices
+    dropped. Yields batches like BucketingSampler."""
+
+    def __init__(self, dataset, batch_size: int, world_size: int = 1,
+                 rank: int = 0):
+        self.n = len(dataset)
+        self.batch_size = batch_size
+        per = (self.n + world_size - 1) // world_size
+        lo = rank * per
+        hi = min(lo + per, self.n)
+        self.indices = list(range(lo, hi))
+
+    def __len__(self) -> int:
+        return (len(self.indices) + self.batch_size - 1) \
+            // self.batch_size
+
+    def __iter__(self):
+        for i in range(0, len(self.indices), self.batch_size):
+            yield self.indices[i:i + self.batch_size]

@@ -132,7 +132,13 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
         vls = AverageMeter()
         hyps, refs = [], []
         with torch.no_grad():
-            vs = BucketingSampler(val_ds, batch_size, 1, 0, seed)
+            # eval shards statically across ranks (reference
+            # StaticDistributedSampler); metrics are rank-local, rank 0
+            # logs its shard
+            from ddlbench_amd.data.translation import (
+                StaticDistributedSampler)
+            vs = StaticDistributedSampler(val_ds, batch_size, world,
+                                          rank)
             for j, batch in enumerate(batches(val_ds, vs)):
                 src, src_len, tgt_in, tgt_out = to_dev(*batch)
                 vls.update(loss_fn(model(src, src_len, tgt_in),

@@ -220,3 +220,16 @@ def test_run_gnmt_real_corpus(tmp_path):
                    device="cpu", log_interval=0, bleu_batches=1,
                    data_dir=str(tmp_path))
     assert res["samples_per_sec"] > 0
+
+
+def test_static_distributed_sampler():
+    from ddlbench_amd.data.translation import (StaticDistributedSampler,
+                                               SyntheticTranslationDataset)
+    ds = SyntheticTranslationDataset(10, 100, max_len=8)
+    shards = [list(StaticDistributedSampler(ds, 2, 3, r))
+              for r in range(3)]
+    flat = sorted(i for shard in shards for b in shard for i in b)
+    assert flat == list(range(10))  # exact cover, no dupes
+    # contiguous, deterministic
+    assert shards[0][0] == [0, 1]
+    assert len(StaticDistributedSampler(ds, 4, 1, 0)) == 3
