@@ -21,7 +21,9 @@ GROUPS = [
     ("ddlbench_bn", lambda n: n.startswith("void bn_") or
      n.startswith("bn_")),
     ("ddlbench_conv", lambda n: "conv_igemm" in n or "conv_wgrad" in n or
-     "dw3x3" in n),
+     "wgrad_combine" in n or "dw3x3" in n),
+    ("ddlbench_pool", lambda n: "maxpool_fwd_nhwc" in n or
+     "maxpool_bwd_nhwc" in n),
     ("ddlbench_sgd_adam", lambda n: "fused_sgd" in n or "fused_adam" in n),
     ("ddlbench_ce_seq", lambda n: n.startswith("void ce_") or
      "revert_varlen" in n or "varlen_mask" in n),
