@@ -137,6 +137,12 @@ class Trainer:
                 self.log.train_step(epoch, cfg.epochs,
                                     int(100.0 * (i + 1) / n_batches),
                                     sps, alloc, reserved, total)
+                if self.dp is not None \
+                        and hasattr(self.dp, "pop_reduce_times"):
+                    rt = self.dp.pop_reduce_times()
+                    if rt:  # DDLB_LOG_REDUCE=1 (extract_reduce_times)
+                        self.log.info("reduce_times_ms: " + " ".join(
+                            f"{v:.3f}" for v in rt))
                 window_start, window_samples = time.perf_counter(), 0
         sync(self.device)
         elapsed = time.perf_counter() - tick

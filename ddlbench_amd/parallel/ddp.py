@@ -24,6 +24,7 @@ Works on the gloo backend for CPU-only multi-process tests.
 from __future__ import annotations
 
 import os
+import time
 from typing import List
 
 import torch
@@ -32,7 +33,8 @@ import torch.nn as nn
 
 
 class _Bucket:
-    __slots__ = ("params", "flat", "views", "pending", "work", "comm")
+    __slots__ = ("params", "flat", "views", "pending", "work", "comm",
+                 "t0", "ev0", "ev1")
 
     def __init__(self) -> None:
         self.params: List[torch.nn.Parameter] = []
@@ -41,11 +43,15 @@ class _Bucket:
         self.pending = 0
         self.work = None
         self.comm = None  # fp32 reduction scratch (DDLB_BUCKET_FP32=1)
+        self.t0 = 0.0     # reduce-time logging (DDLB_LOG_REDUCE=1)
+        self.ev0 = None
+        self.ev1 = None
 
 
 class BucketedDataParallel(nn.Module):
     def __init__(self, module: nn.Module, bucket_mb: float = 0.0,
-                 process_group=None, average: bool = True):
+                 process_group=None, average: bool = True,
+                 log_reduce_times: bool = False):
         super().__init__()
         self.module = module
         self.pg = process_group
@@ -54,6 +60,12 @@ class BucketedDataParallel(nn.Module):
             bucket_mb = float(os.environ.get("DDLB_BUCKET_MB", "32"))
         self.bucket_bytes = int(bucket_mb * 2**20)
         self._fp32_reduce = os.environ.get("DDLB_BUCKET_FP32", "0") == "1"
+        # per-bucket all-reduce span logging (the reference extracts the
+        # same curves from a patched DDP's log_reduce_times stdout,
+        # profiler utils/all_reduce/extract_reduce_times.py:7-30)
+        self._log_reduce = log_reduce_times or \
+            os.environ.get("DDLB_LOG_REDUCE", "0") == "1"
+        self._reduce_ms: List[float] = []
         self.world_size = (dist.get_world_size(self.pg)
                            if dist.is_initialized() else 1)
         self._buckets: List[_Bucket] = []
@@ -135,6 +147,12 @@ class BucketedDataParallel(nn.Module):
             # the 7-link xGMI mesh. The reference's horovod reduced
             # fp32; DDLB_BUCKET_FP32=1 reproduces that numerics at 2x
             # the wire bytes (docs/MULTIGPU.md).
+            if self._log_reduce:
+                if b.flat.is_cuda:
+                    b.ev0 = torch.cuda.Event(enable_timing=True)
+                    b.ev0.record()
+                else:
+                    b.t0 = time.perf_counter()
             if self._fp32_reduce and b.flat.dtype != torch.float32:
                 b.comm = b.flat.to(torch.float32)
                 b.work = dist.all_reduce(b.comm, op=dist.ReduceOp.SUM,
@@ -156,10 +174,20 @@ class BucketedDataParallel(nn.Module):
         imagenet_horovod.py:36)."""
         if self.world_size <= 1:
             return
+        ev_pairs = []
         for b in self._buckets:
             if b.work is not None:
                 b.work.wait()
                 b.work = None
+                if self._log_reduce:
+                    if b.ev0 is not None:
+                        b.ev1 = torch.cuda.Event(enable_timing=True)
+                        b.ev1.record()
+                        ev_pairs.append((b.ev0, b.ev1))
+                        b.ev0 = None
+                    else:
+                        self._reduce_ms.append(
+                            (time.perf_counter() - b.t0) * 1e3)
                 if b.comm is not None:
                     if self.average:
                         b.comm.div_(self.world_size)
@@ -171,6 +199,18 @@ class BucketedDataParallel(nn.Module):
             for p in b.params:
                 if p.grad is None:
                     p.grad = b.views[p]
+        if ev_pairs:
+            torch.cuda.synchronize()
+            self._reduce_ms.extend(e0.elapsed_time(e1)
+                                   for e0, e1 in ev_pairs)
+
+    def pop_reduce_times(self) -> List[float]:
+        """Drain logged per-bucket all-reduce spans (ms), bucket-ready →
+        collective-complete as seen by the compute stream — overlap with
+        the rest of backward included. Empty unless DDLB_LOG_REDUCE=1 or
+        log_reduce_times=True."""
+        out, self._reduce_ms = self._reduce_ms, []
+        return out
 
     def zero_grad_buckets(self) -> None:
         if self.world_size <= 1:

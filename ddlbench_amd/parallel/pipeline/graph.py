@@ -188,6 +188,49 @@ class Graph:
         lines.append("}")
         return "\n".join(lines) + "\n"
 
+    def plot_cdf(self, path: str) -> None:
+        """CDF of cumulative compute time over layers in topo order (the
+        reference's plot_cdfs, graph.py:482-616). PNG/PDF by extension."""
+        import matplotlib
+        matplotlib.use("Agg")
+        import matplotlib.pyplot as plt
+        nodes = self.topological_sort()
+        total = sum(n.compute_time for n in nodes) or 1.0
+        xs = list(range(1, len(nodes) + 1))
+        acc, ys = 0.0, []
+        for n in nodes:
+            acc += n.compute_time
+            ys.append(acc / total)
+        fig, ax = plt.subplots(figsize=(6, 4))
+        ax.step(xs, ys, where="post")
+        ax.set_xlabel("layer (topological order)")
+        ax.set_ylabel("cumulative compute time fraction")
+        ax.set_ylim(0, 1.02)
+        ax.grid(True, alpha=0.3)
+        fig.tight_layout()
+        fig.savefig(path)
+        plt.close(fig)
+
+    def plot_bars(self, path: str) -> None:
+        """Per-layer fwd/bwd stacked time bars (the reference's
+        plot_bar_graph, graph.py:482-616)."""
+        import matplotlib
+        matplotlib.use("Agg")
+        import matplotlib.pyplot as plt
+        nodes = self.topological_sort()
+        xs = list(range(len(nodes)))
+        fwd = [n.fwd_time * 1e3 for n in nodes]
+        bwd = [n.bwd_time * 1e3 for n in nodes]
+        fig, ax = plt.subplots(figsize=(max(6, len(nodes) * 0.12), 4))
+        ax.bar(xs, fwd, label="forward")
+        ax.bar(xs, bwd, bottom=fwd, label="backward")
+        ax.set_xlabel("layer (topological order)")
+        ax.set_ylabel("time (ms)")
+        ax.legend()
+        fig.tight_layout()
+        fig.savefig(path)
+        plt.close(fig)
+
     # ---- serialization (our own line format) ----------------------------
     def dumps(self) -> str:
         lines = []

@@ -55,10 +55,13 @@ class PartitionResult:
 
     @property
     def module_to_stage_map(self) -> List[int]:
+        # negative ids are synthetic (the profiler's Input node) — not
+        # modules, so they never appear in the map
         m = {}
         for si, st in enumerate(self.stages):
             for l in st.layers:
-                m[l] = si
+                if l >= 0:
+                    m[l] = si
         return [m[i] for i in sorted(m)]
 
     @property
@@ -84,6 +87,21 @@ def _dp_allreduce_time(r: int, param_bytes: float, bw: float) -> float:
     if r <= 1:
         return 0.0
     return 4.0 * (r - 1) * param_bytes / (r * bw)
+
+
+def zero_input_nodes(graph: Graph) -> int:
+    """Zero compute/params on synthetic Input nodes before partitioning.
+
+    The reference's optimizer does the same so data-loading time (carried
+    on the profiler's appended Input node) never skews stage placement
+    (optimizer_graph_hierarchical.py:193-213). Returns the count."""
+    n = 0
+    for nd in graph.nodes.values():
+        if nd.desc == "Input" or nd.desc.startswith("Input("):
+            nd.fwd_time = nd.bwd_time = 0.0
+            nd.parameter_size = 0.0
+            n += 1
+    return n
 
 
 def partition_chain(graph: Graph, num_gpus: int, bw: float = XGMI_BW,
@@ -284,6 +302,7 @@ def partition_graph(graph: Graph, num_gpus: int,
     super-nodes before partitioning (graph.py:139-227,
     compress_graph_branches.py) so the antichain enumeration stays
     tractable; stage ids are propagated back to the original nodes."""
+    zero_input_nodes(graph)
     if graph.is_chain():
         return partition_chain(graph, num_gpus, **kw)
     if compress:
@@ -302,8 +321,7 @@ def partition_graph(graph: Graph, num_gpus: int,
                 stages.append(Stage(layers=layers, replicas=st.replicas,
                                     time=st.time))
                 for l in layers:
-                    graph.nodes[l].stage_id = cg.nodes[
-                        st.layers[0]].stage_id if False else                         len(stages) - 1
+                    graph.nodes[l].stage_id = len(stages) - 1
             return PartitionResult(stages=stages,
                                    bottleneck=res.bottleneck,
                                    pure_dp_time=res.pure_dp_time,

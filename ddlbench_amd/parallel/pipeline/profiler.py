@@ -144,6 +144,42 @@ def profile_sequential(seq: nn.Sequential, sample: torch.Tensor,
     return Graph.chain(nodes)
 
 
+def measure_data_time(loader, iters: int = 10) -> float:
+    """Average seconds per batch fetched from `loader` (the reference
+    times data loading the same way and carries it on the Input node,
+    profiler main.py:402-407)."""
+    it = iter(loader)
+    times = []
+    for _ in range(iters):
+        t0 = time.perf_counter()
+        try:
+            next(it)
+        except StopIteration:
+            break
+        times.append(time.perf_counter() - t0)
+    return sum(times) / len(times) if times else 0.0
+
+
+def append_input_node(g: Graph, data_time: float = 0.0,
+                      activation_bytes: float = 0.0) -> Node:
+    """Prepend a synthetic "Input" node carrying the data-loading time.
+
+    Mirrors the reference profiler's appended Input node (profiler
+    main.py:402-407). The partitioner zeroes it before the DP
+    (partition.zero_input_nodes) so data time never skews placement but
+    stays visible in the serialized graph/plots. The node id is
+    min(ids)-1, keeping every module_to_stage_map index untouched."""
+    srcs = [n.node_id for n in g.sources()]
+    nid = (min(g.nodes) - 1) if g.nodes else 0
+    node = g.add_node(Node(nid, desc=f"Input({data_time * 1e3:.3f} ms)"
+                           if data_time else "Input",
+                           fwd_time=data_time,
+                           activation_size=float(activation_bytes)))
+    for s in srcs:
+        g.add_edge(nid, s)
+    return node
+
+
 def profile_module_graph(model: nn.Module, *sample_args,
                          device: Optional[torch.device] = None,
                          iters: int = 6, warmup: int = 2) -> Graph:

@@ -20,6 +20,19 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
 from ddlbench_amd.utils import parse_result_line  # noqa: E402
 
 
+def extract_reduce_times(path: str):
+    """Per-bucket all-reduce spans (ms) from `reduce_times_ms:` lines
+    (DDLB_LOG_REDUCE=1) — the counterpart of the reference's
+    profiler utils/all_reduce/extract_reduce_times.py:7-30."""
+    times = []
+    with open(path) as f:
+        for line in f:
+            if "reduce_times_ms:" in line:
+                vals = line.split("reduce_times_ms:", 1)[1].split()
+                times.extend(float(v) for v in vals)
+    return times
+
+
 def parse_log(path: str) -> dict:
     out = {"train": [], "epochs": [], "final": None}
     with open(path) as f:
@@ -34,6 +47,9 @@ def parse_log(path: str) -> dict:
                 out["epochs"].append(d)
             else:
                 out["final"] = d
+    reduce_ms = extract_reduce_times(path)
+    if reduce_ms:
+        out["reduce_times_ms"] = reduce_ms
     return out
 
 

@@ -170,3 +170,35 @@ def _worker_channels_last(rank, world, port):
 def test_channels_last_buckets_two_ranks(free_port):
     mp.spawn(_worker_channels_last, args=(2, free_port), nprocs=2,
              join=True)
+
+
+def _worker_reduce_log(rank, world, port):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from ddlbench_amd.parallel import BucketedDataParallel
+
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(4, 64), torch.nn.ReLU(),
+                                torch.nn.Linear(64, 2))
+    dp = BucketedDataParallel(model, bucket_mb=0.0001,
+                              log_reduce_times=True)
+    nb = len(dp._buckets)
+    assert nb > 1
+    dp.zero_grad_buckets()
+    dp(torch.randn(8, 4)).sum().backward()
+    dp.finalize_backward()
+    times = dp.pop_reduce_times()
+    assert len(times) == nb
+    assert all(t >= 0.0 for t in times)
+    assert dp.pop_reduce_times() == []  # drained
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_reduce_time_logging_two_ranks(free_port):
+    """Per-bucket all-reduce spans logged and drained (the reference's
+    extract_reduce_times pipeline, utils/all_reduce/)."""
+    mp.spawn(_worker_reduce_log, args=(2, free_port), nprocs=2, join=True)

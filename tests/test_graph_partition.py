@@ -279,3 +279,42 @@ def test_compress_branches_inception():
     for src, dsts in g.edges.items():
         for dst in dsts:
             assert g.nodes[src].stage_id <= g.nodes[dst].stage_id
+
+
+def test_input_node_appended_and_zeroed():
+    """The profiler's synthetic Input node carries data-loading time
+    (reference profiler main.py:402-407); the partitioner zeroes it
+    (optimizer_graph_hierarchical.py:193-213) so placement and the
+    module map are unchanged."""
+    from ddlbench_amd.parallel.pipeline.partition import partition_graph
+    from ddlbench_amd.parallel.pipeline.profiler import append_input_node
+    base = _chain([1.0, 1.0, 1.0, 1.0])
+    res0 = partition_graph(base, 2, straight=True)
+
+    g = _chain([1.0, 1.0, 1.0, 1.0])
+    node = append_input_node(g, data_time=100.0, activation_bytes=512.0)
+    assert node.node_id == -1
+    assert g.is_chain()
+    assert [n.node_id for n in g.topological_sort()][0] == -1
+    res = partition_graph(g, 2, straight=True)
+    # huge data time must not skew the bottleneck or the module map
+    assert res.bottleneck == pytest.approx(res0.bottleneck)
+    assert res.module_to_stage_map == res0.module_to_stage_map
+    assert g.nodes[-1].fwd_time == 0.0  # zeroed in place
+    assert g.nodes[-1].stage_id == 0
+
+
+def test_measure_data_time():
+    from ddlbench_amd.parallel.pipeline.profiler import measure_data_time
+    t = measure_data_time([torch.zeros(1)] * 4, iters=3)
+    assert isinstance(t, float) and t >= 0.0
+    assert measure_data_time([], iters=3) == 0.0
+
+
+def test_plot_cdf_and_bars(tmp_path):
+    g = _chain([1.0, 2.0, 3.0])
+    cdf, bars = tmp_path / "cdf.png", tmp_path / "bars.png"
+    g.plot_cdf(str(cdf))
+    g.plot_bars(str(bars))
+    assert cdf.stat().st_size > 0
+    assert bars.stat().st_size > 0

@@ -113,3 +113,15 @@ def test_conv_mfma_module_cpu_fallback():
     assert convert_convs(model) == 1
     assert isinstance(model[0], nn.Conv2d)
     assert isinstance(model[1], Conv2dMFMA)
+
+
+def test_extract_reduce_times(tmp_path):
+    sys.path.insert(0, "run")
+    from process_output import extract_reduce_times
+    log = tmp_path / "run.log"
+    log.write_text(
+        "train | 1/3 epoch (50%) | 100.000 samples/sec (estimated) | "
+        "mem (GB): 1.000 (2.000) / 288.000\n"
+        "reduce_times_ms: 1.250 0.750\n"
+        "reduce_times_ms: 2.000\n")
+    assert extract_reduce_times(str(log)) == [1.25, 0.75, 2.0]
