@@ -20,6 +20,9 @@ if __name__ == "__main__":
     p.add_argument("--dtype", default="float32",
                    choices=["float32", "bfloat16"])
     p.add_argument("--device", default="auto")
+    p.add_argument("--data-dir", default=os.environ.get("DATADIR", ""),
+                   help="parallel corpus root (train.src/train.tgt); "
+                        "synthetic token stream when empty")
     a = p.parse_args()
     run_gnmt_pipeline(
         epochs=int(os.environ.get("EPOCHS", 3)),
@@ -27,4 +30,5 @@ if __name__ == "__main__":
         n_minibatches=a.minibatches, vocab=a.vocab, hidden=a.hidden,
         layers=a.layers, lr=a.lr, dtype=a.dtype, device=a.device,
         src_len_max=a.src_len, tgt_len=a.tgt_len,
-        log_interval=int(os.environ.get("LOGINTER", 0)))
+        log_interval=int(os.environ.get("LOGINTER", 0)),
+        data_dir=a.data_dir)

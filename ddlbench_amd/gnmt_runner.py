@@ -166,11 +166,19 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
 def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
                       vocab=32320, hidden=1024, layers=4, lr=2.5e-4,
                       dtype="float32", device="auto", seed=42,
-                      src_len_max=48, tgt_len=48, log_interval=0) -> dict:
+                      src_len_max=48, tgt_len=48, log_interval=0,
+                      data_dir="") -> dict:
     """GNMT through the 1F1B pipeline — the reference's
     translation/main_with_runtime.py flow on the
     tuple-I/O StageRuntime. Straight pipeline, one stage per rank;
-    fixed (padded) sequence lengths give static edge shapes."""
+    fixed (padded) sequence lengths give static edge shapes.
+
+    data_dir: parallel corpus root (<dir>/train.{src,tgt}) — batches
+    are padded/truncated to the fixed (src_len_max, tgt_len) shapes the
+    static pipeline edges need; every rank builds the same dataset and
+    epoch permutation, so the first stage's inputs and the last stage's
+    targets line up without a data channel (the reference ships the
+    target through the pipeline instead; runtime.py:540-543)."""
     import torch.distributed as dist
 
     from ddlbench_amd.models.gnmt import (GNMT, LabelSmoothingLoss,
@@ -194,6 +202,13 @@ def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
     cfg.device = device
     dev = resolve_device(cfg, env.local_rank)
     dt = torch.bfloat16 if dtype == "bfloat16" else torch.float32
+
+    ds = None
+    if data_dir:
+        from ddlbench_amd.data.tokenizer import TextTranslationDataset
+        ds = TextTranslationDataset(
+            data_dir, "train", max_len=max(src_len_max, tgt_len + 1))
+        vocab = max(len(ds.tok_src), len(ds.tok_tgt), 8)
 
     torch.manual_seed(seed)
     model = GNMT(vocab_size=vocab, hidden_size=hidden, num_layers=layers,
@@ -268,6 +283,36 @@ def run_gnmt_pipeline(epochs=1, batch_size=32, n_minibatches=16,
         versioned=plan.num_warmup(stage) > 0)
 
     def providers(epoch):
+        if ds is not None:
+            # real corpus: identical epoch permutation on every rank,
+            # fixed-shape padding (PAD=0) for the static edges
+            g = torch.Generator().manual_seed(seed * 131 + epoch * 7919)
+            perm = torch.randperm(len(ds), generator=g).tolist()
+
+            def fixed(mb):
+                start = (mb * batch_size) % len(ds)
+                src = torch.zeros(src_len_max, batch_size,
+                                  dtype=torch.long)
+                src_len = torch.ones(batch_size, dtype=torch.long)
+                tgt_full = torch.zeros(tgt_len + 1, batch_size,
+                                       dtype=torch.long)
+                for b in range(batch_size):
+                    s, t = ds[perm[(start + b) % len(ds)]]
+                    s, t = s[:src_len_max], t[:tgt_len + 1]
+                    src[:len(s), b] = s
+                    src_len[b] = len(s)
+                    tgt_full[:len(t), b] = t
+                return src, src_len, tgt_full
+
+            def input_provider(mb):
+                src, src_len, tf = fixed(mb)
+                return src, src_len, tf[:-1]
+
+            def target_provider(mb):
+                return fixed(mb)[2][1:]
+
+            return input_provider, target_provider
+
         def tgt_full(mb):
             g = torch.Generator().manual_seed(
                 seed * 131 + epoch * 7919 + mb)
