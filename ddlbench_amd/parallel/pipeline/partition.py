@@ -82,6 +82,25 @@ class PartitionResult:
         with open(path, "w") as f:
             json.dump(self.to_conf(), f, indent=2)
 
+    def describe(self) -> str:
+        """Human-readable split analysis — the reference prints the
+        same `(split_start, split_end) time replication_factor` walk
+        plus predicted pipeline-vs-DP throughput
+        (optimizer_graph_hierarchical.py:169-191, 348-374); here it is
+        a return value, not bash-parsed stdout."""
+        lines = [f"{self.num_gpus} GPUs, {len(self.stages)} stage(s):"]
+        for si, st in enumerate(self.stages):
+            lo, hi = min(st.layers), max(st.layers)
+            lines.append(
+                f"  stage {si}: layers ({lo}, {hi})  "
+                f"time {st.time * 1e3:.3f} ms  replicas {st.replicas}")
+        lines.append(
+            f"  pipeline bottleneck {self.bottleneck * 1e3:.3f} ms"
+            f" vs pure-DP {self.pure_dp_time * 1e3:.3f} ms"
+            f" (speedup {self.pure_dp_time / self.bottleneck:.2f}x)"
+            if self.bottleneck > 0 else "  (zero-time pipeline)")
+        return "\n".join(lines)
+
 
 def _dp_allreduce_time(r: int, param_bytes: float, bw: float) -> float:
     if r <= 1:

@@ -54,6 +54,8 @@ def _make_plan(cfg: BenchConfig, seq, device, world: int):
         os.makedirs(prof_dir, exist_ok=True)
         graph.save(os.path.join(prof_dir, "graph.txt"))
         result.save(os.path.join(prof_dir, "conf.json"))
+        print(result.describe(), flush=True)  # the reference's split
+        # analysis stdout (optimizer_graph_hierarchical.py:169-191)
         payload = [{
             "module_to_stage_map": result.module_to_stage_map,
             "replicas": [s.replicas for s in result.stages],

@@ -318,3 +318,14 @@ def test_plot_cdf_and_bars(tmp_path):
     g.plot_bars(str(bars))
     assert cdf.stat().st_size > 0
     assert bars.stat().st_size > 0
+
+
+def test_partition_describe():
+    """Reference-style split analysis text
+    (optimizer_graph_hierarchical.py:169-191 stdout analogue)."""
+    g = _chain([1.0] * 4)
+    res = partition_chain(g, 2, straight=True)
+    txt = res.describe()
+    assert "2 GPUs" in txt and "stage 0" in txt and "stage 1" in txt
+    assert "replicas 1" in txt
+    assert "vs pure-DP" in txt
