@@ -71,3 +71,32 @@ def test_run_sh_real_data_trains_from_tree(tmp_path):
         cwd=".")
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     assert "valid accuracy:" in out.stdout
+
+
+def test_run_sh_gpipe_cpu(tmp_path):
+    """gpipe front-end path (single process; 1 partition on CPU)."""
+    import os
+    env = dict(os.environ)
+    env.update(HOME=str(tmp_path))
+    out = subprocess.run(
+        ["bash", "run/run.sh", "-b", "mnist", "-f", "gpipe",
+         "-m", "resnet18", "-e", "1", "-B", "8", "-M", "2", "-p", "0"],
+        capture_output=True, text=True, timeout=900, env=env)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "valid accuracy:" in out.stdout
+
+
+def test_run_sh_pipedream_two_ranks_cpu(tmp_path):
+    """pipedream front-end: profile -> partition (analysis printed) ->
+    1F1B runtime, 2 gloo ranks on CPU."""
+    import os
+    env = dict(os.environ)
+    env.update(HOME=str(tmp_path), MASTER_PORT="29551")
+    out = subprocess.run(
+        ["bash", "run/run.sh", "-b", "mnist", "-f", "pipedream",
+         "-g", "2", "-m", "resnet18", "-e", "1", "-B", "8", "-M", "4",
+         "-p", "0"],
+        capture_output=True, text=True, timeout=900, env=env)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "valid accuracy:" in out.stdout
+    assert "2 GPUs" in out.stdout  # the partition analysis line
