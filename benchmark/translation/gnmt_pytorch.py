@@ -22,6 +22,10 @@ def main(ddp=False):
     p.add_argument("--data-dir", default=os.environ.get("DATADIR", ""),
                    help="parallel corpus root (train.src/train.tgt); "
                         "synthetic token stream when empty")
+    p.add_argument("--optimizer", default="sgd",
+                   choices=["sgd", "adam"],
+                   help="adam = the reference GNMT optimizer "
+                        "(fused multi-tensor step either way)")
     a = p.parse_args()
     run_gnmt(epochs=int(os.environ.get("EPOCHS", 3)),
              batch_size=int(os.environ.get("BATCH_SIZE", 64)),
@@ -29,7 +33,7 @@ def main(ddp=False):
              dataset_size=a.dataset_size, vocab=a.vocab, hidden=a.hidden,
              layers=a.layers, lr=a.lr, dtype=a.dtype, device=a.device,
              max_len=a.max_len, ddp=ddp, kernel_backend=a.kernel_backend,
-             data_dir=a.data_dir)
+             data_dir=a.data_dir, optimizer=a.optimizer)
 
 if __name__ == "__main__":
     main(ddp=False)

@@ -25,7 +25,7 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
              hidden=1024, layers=4, lr=0.25e-3, dtype="float32",
              device="auto", log_interval=25, seed=42, max_len=50,
              ddp=False, kernel_backend="auto", bleu_batches=2,
-             data_dir="") -> dict:
+             data_dir="", optimizer="sgd") -> dict:
     from ddlbench_amd.parallel import (BucketedDataParallel,
                                        allreduce_mean_scalar,
                                        init_distributed)
@@ -68,10 +68,16 @@ def run_gnmt(epochs=3, batch_size=64, dataset_size=2000, vocab=32320,
     if dt != torch.float32:
         model = model.to(dt)
     dp = BucketedDataParallel(model) if world > 1 else None
-    # Adam is the reference GNMT optimizer; SGD keeps the fused path —
-    # throughput benchmarking is optimizer-agnostic, use fused SGD
-    opt = FusedSGD(model.parameters(), lr=lr * world, momentum=0.9,
-                   backend=kernel_backend)
+    # Adam is the reference GNMT optimizer (runtime/translation
+    # main_with_runtime.py position); both run the fused multi-tensor
+    # step — SGD stays the throughput default
+    if optimizer == "adam":
+        from ddlbench_amd.ops.adam import FusedAdam
+        opt = FusedAdam(model.parameters(), lr=lr * world,
+                        backend=kernel_backend)
+    else:
+        opt = FusedSGD(model.parameters(), lr=lr * world, momentum=0.9,
+                       backend=kernel_backend)
     loss_fn = LabelSmoothingLoss(0.1)
 
     sampler = BucketingSampler(train_ds, batch_size, world, rank, seed)

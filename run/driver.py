@@ -36,7 +36,7 @@ DEFAULTS = dict(model="resnet50", gpus=1, epochs=3, batch_size=0,
                 dtype="float32", log_interval=25, synthetic_scale=0.01,
                 output_dir="out/driver", master_port=29531,
                 extra_args=[])
-VALID_BM = ("mnist", "cifar10", "imagenet", "highres")
+VALID_BM = ("mnist", "cifar10", "imagenet", "highres", "translation")
 VALID_FW = ("pytorch", "horovod", "gpipe", "pipedream")
 
 
@@ -50,6 +50,10 @@ def load_config(path: str) -> dict:
         raise ValueError(f"benchmark must be one of {VALID_BM}")
     if cfg["framework"] not in VALID_FW:
         raise ValueError(f"framework must be one of {VALID_FW}")
+    if cfg["benchmark"] == "translation" and cfg["framework"] == "gpipe":
+        # GNMT runs single / DP / 1F1B (the reference ran it only
+        # through the pipedream driver)
+        raise ValueError("translation supports pytorch|horovod|pipedream")
     out = dict(DEFAULTS)
     out.update(cfg)
     return out
@@ -57,6 +61,21 @@ def load_config(path: str) -> dict:
 
 def build_command(cfg: dict):
     ds = cfg["benchmark"]
+    if ds == "translation":
+        # GNMT: yml-config launch is the reference's entry for this
+        # workload (runtime/translation/driver_configs/*.yml)
+        script = os.path.join(ROOT, "benchmark", "translation",
+                              f"gnmt_{cfg['framework']}.py")
+        args = ["--dtype", cfg["dtype"]]
+        args += list(map(str, cfg["extra_args"]))
+        if cfg["framework"] == "pipedream" and cfg["gpus"] > 1 or \
+                cfg["framework"] == "horovod" and cfg["gpus"] > 1:
+            return [sys.executable, "-m", "torch.distributed.run",
+                    "--nnodes=1", f"--nproc-per-node={cfg['gpus']}",
+                    "--master-addr", "127.0.0.1",
+                    "--master-port", str(cfg["master_port"]),
+                    script] + args
+        return [sys.executable, script] + args
     script_ds = "imagenet" if ds == "highres" else ds
     script = os.path.join(ROOT, "benchmark", script_ds,
                           f"{script_ds}_{cfg['framework']}.py")

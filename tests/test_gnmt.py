@@ -233,3 +233,12 @@ def test_static_distributed_sampler():
     # contiguous, deterministic
     assert shards[0][0] == [0, 1]
     assert len(StaticDistributedSampler(ds, 4, 1, 0)) == 3
+
+
+def test_gnmt_runner_adam():
+    """optimizer="adam" — the reference GNMT optimizer — trains."""
+    from ddlbench_amd.gnmt_runner import run_gnmt
+    res = run_gnmt(epochs=1, batch_size=4, dataset_size=8, vocab=48,
+                   hidden=16, layers=2, device="cpu", log_interval=0,
+                   bleu_batches=1, optimizer="adam")
+    assert res["samples_per_sec"] > 0
